@@ -1,0 +1,49 @@
+"""Model zoo.
+
+Reference zoo: MLP (MNIST), CCT (CIFAR-10).  BASELINE.json
+additionally requires ResNet-18 and WideResNet-28-10 — provided here.
+
+``get_model(name, **kw)`` is the string registry used by the CLI and bench.
+"""
+from __future__ import annotations
+
+from torch import nn
+
+from .mlp import MLP
+from .resnet import ResNet, WideResNet, resnet18, wide_resnet28_10
+from .cct import CCT, CCTNet, cct_2_3x2_32
+
+_REGISTRY = {
+    "mlp": lambda **kw: MLP(**kw),
+    "resnet18": lambda **kw: resnet18(**kw),
+    "wrn28_10": lambda **kw: wide_resnet28_10(**kw),
+    "wideresnet28_10": lambda **kw: wide_resnet28_10(**kw),
+    "cct": lambda **kw: cct_2_3x2_32(**kw),
+    "cct_2_3x2_32": lambda **kw: cct_2_3x2_32(**kw),
+}
+
+
+def get_model(name: str, **kw) -> nn.Module:
+    try:
+        factory = _REGISTRY[name.lower()]
+    except KeyError:
+        raise KeyError(f"unknown model {name!r}; available: {sorted(_REGISTRY)}")
+    return factory(**kw)
+
+
+def register_model(name: str):
+    def deco(fn):
+        _REGISTRY[name.lower()] = fn
+        return fn
+    return deco
+
+
+def num_params(model: nn.Module) -> int:
+    return sum(p.numel() for p in model.parameters() if p.requires_grad)
+
+
+__all__ = [
+    "MLP", "ResNet", "WideResNet", "CCT", "CCTNet",
+    "resnet18", "wide_resnet28_10", "cct_2_3x2_32",
+    "get_model", "register_model", "num_params",
+]

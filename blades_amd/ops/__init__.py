@@ -1,0 +1,196 @@
+"""Op dispatch: CDNA4 HIP kernels on GPU, pure-torch reference on CPU.
+
+Policy (fail-loudly contract):
+
+* CPU tensor            -> torch reference implementation (torch_ref.py).
+* GPU tensor, ext built -> hand-written HIP kernel (csrc/).
+* GPU tensor, no ext    -> RuntimeError.  A GPU run silently falling back
+  to eager PyTorch would invalidate every benchmark claim, so it refuses.
+* ``BLADES_AMD_FORCE_TORCH=1`` forces the torch path everywhere (used by
+  the GPU numerics tests to A/B kernel vs reference on device).
+
+Ops whose torch form is already a single fused ROCm kernel (flat axpy,
+nan_to_num, RNG fill) stay on torch on both devices; everything listed in
+SURVEY.md §2.4 with real fusion opportunity has a HIP implementation.
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from . import torch_ref as _ref
+from .torch_ref import (  # ops that are torch on both devices
+    col_var,
+    clip_to_norm,
+    row_clip_to_norm_,
+    krum_scores,
+    philox_normal,
+    flat_sgd_step_,
+    sanitize_,
+)
+
+Tensor = torch.Tensor
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_extension():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from blades_amd import _hip_ops  # built in-tree by setup.py / __graft_entry__
+        _EXT = _hip_ops
+    except ImportError as e:
+        _EXT_ERR = str(e)
+        _EXT = None
+    return _EXT
+
+
+def hip_available() -> bool:
+    return _load_extension() is not None
+
+
+def _force_torch() -> bool:
+    return os.environ.get("BLADES_AMD_FORCE_TORCH", "0") == "1"
+
+
+def _route(U: Tensor):
+    """Return the HIP extension if this tensor must run on it, else None."""
+    if U.is_cuda and not _force_torch():
+        ext = _load_extension()
+        if ext is None:
+            raise RuntimeError(
+                "blades_amd HIP extension not built but a GPU tensor hit the "
+                f"op layer (import error: {_EXT_ERR}). Build it with "
+                "`python setup.py build_ext --inplace` "
+                "(PYTORCH_ROCM_ARCH=gfx950); refusing to fall back to eager "
+                "PyTorch on GPU."
+            )
+        return ext
+    return None
+
+
+# --------------------------------------------------------------- column ops
+
+def col_mean(U: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.col_mean(U.contiguous())
+    return _ref.col_mean(U)
+
+
+def col_median(U: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.col_median(U.contiguous())
+    return _ref.col_median(U)
+
+
+def trimmed_mean(U: Tensor, b: int) -> Tensor:
+    K = U.shape[0]
+    if K - 2 * b <= 0:
+        raise ValueError(f"trimmed_mean needs K > 2b (K={K}, b={b})")
+    ext = _route(U)
+    if ext is not None:
+        return ext.trimmed_mean(U.contiguous(), b)
+    return _ref.trimmed_mean(U, b)
+
+
+def weighted_col_sum(U: Tensor, w: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.weighted_col_sum(U.contiguous(), w.to(U).contiguous())
+    return _ref.weighted_col_sum(U, w)
+
+
+def masked_col_mean(U: Tensor, mask: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.masked_col_mean(U.contiguous(), mask.to(torch.bool).contiguous())
+    return _ref.masked_col_mean(U, mask)
+
+
+def masked_col_mean_std(U: Tensor, mask: Tensor,
+                        unbiased: bool = True) -> Tuple[Tensor, Tensor]:
+    ext = _route(U)
+    if ext is not None:
+        return ext.masked_col_mean_std(U.contiguous(),
+                                       mask.to(torch.bool).contiguous(), unbiased)
+    return _ref.masked_col_mean_std(U, mask, unbiased)
+
+
+# ------------------------------------------------------------------ row ops
+
+def row_sq_norms(U: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.row_sq_norms(U.contiguous())
+    return _ref.row_sq_norms(U)
+
+
+def row_norms(U: Tensor) -> Tensor:
+    return row_sq_norms(U).sqrt()
+
+
+def row_diff_norms(U: Tensor, z: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.row_diff_sq_norms(U.contiguous(), z.contiguous()).sqrt()
+    return _ref.row_diff_norms(U, z)
+
+
+def row_dots(U: Tensor, v: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.row_dots(U.contiguous(), v.contiguous())
+    return _ref.row_dots(U, v)
+
+
+def pairwise_sq_dists(U: Tensor) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        G = ext.gram(U.contiguous())
+        sq = G.diagonal()
+        D = sq.unsqueeze(0) + sq.unsqueeze(1) - 2 * G
+        D.fill_diagonal_(0)
+        return D.clamp_min_(0)
+    return _ref.pairwise_sq_dists(U)
+
+
+def gram(U: Tensor) -> Tensor:
+    """U @ U.T — MFMA f32 on GPU."""
+    ext = _route(U)
+    if ext is not None:
+        return ext.gram(U.contiguous())
+    return U @ U.t()
+
+
+def cos_sim_gram(U: Tensor, eps: float = 1e-8) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        norms = ext.row_sq_norms(U.contiguous()).sqrt().clamp_min(eps)
+        G = ext.gram(U.contiguous())
+        G = G / norms.unsqueeze(0) / norms.unsqueeze(1)
+        return G.clamp_(-1.0, 1.0)
+    return _ref.cos_sim_gram(U, eps)
+
+
+def centered_clip_iter(U: Tensor, v: Tensor, tau: float) -> Tensor:
+    ext = _route(U)
+    if ext is not None:
+        return ext.centered_clip_iter(U.contiguous(), v.contiguous(), float(tau))
+    return _ref.centered_clip_iter(U, v, tau)
+
+
+__all__ = [
+    "col_mean", "col_median", "trimmed_mean", "col_var",
+    "weighted_col_sum", "masked_col_mean", "masked_col_mean_std",
+    "row_sq_norms", "row_norms", "row_diff_norms", "row_dots",
+    "pairwise_sq_dists", "gram", "cos_sim_gram",
+    "krum_scores", "clip_to_norm", "centered_clip_iter", "row_clip_to_norm_",
+    "philox_normal", "flat_sgd_step_", "sanitize_", "hip_available",
+]

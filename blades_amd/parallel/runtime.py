@@ -1,0 +1,126 @@
+"""Distributed runtime: one process per MI355X, RCCL over xGMI.
+
+Replaces the reference's Ray actor pool + object store (reference:
+simulator.py:90-98, actor.py:6-48).  Mapping:
+
+* model broadcast (pickled nn.Module per round)  -> one ncclBroadcast of the
+  flat θ vector at run start; afterwards every rank applies the same
+  deterministic aggregate, so θ stays replicated with ZERO per-round
+  broadcast traffic;
+* update gather (Ray futures of CPU tensors)     -> ncclAllGather of the
+  rank-local [K/ws, d] update slab (device-resident, no host copies).
+  All-gather (not gather-to-root) so omniscient attacks and aggregation run
+  rank-local with full knowledge — no second broadcast;
+* eval metric gather                             -> all_gather_object of the
+  small per-client metric dicts.
+
+Initialization follows torchrun env (RANK/WORLD_SIZE/LOCAL_RANK,
+MASTER_ADDR/PORT); backend "nccl" IS RCCL on ROCm, "gloo" drives the same
+code paths on CPU CI.  world_size == 1 short-circuits every collective.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+
+class DistributedRuntime:
+    def __init__(self, device: Optional[torch.device] = None,
+                 timeout_s: float = 600.0):
+        self.rank = int(os.environ.get("RANK", "0"))
+        self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
+        self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
+
+        if self.world_size > 1 and not dist.is_initialized():
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            dist.init_process_group(
+                backend=backend,
+                timeout=datetime.timedelta(seconds=timeout_s),
+            )
+
+        if device is not None:
+            self.device = torch.device(device)
+        elif torch.cuda.is_available():
+            self.device = torch.device(f"cuda:{self.local_rank}")
+            torch.cuda.set_device(self.device)
+        else:
+            self.device = torch.device("cpu")
+
+    # ------------------------------------------------------------- helpers
+    @property
+    def distributed(self) -> bool:
+        return self.world_size > 1
+
+    def is_main(self) -> bool:
+        return self.rank == 0
+
+    def barrier(self) -> None:
+        if self.distributed:
+            dist.barrier()
+
+    def shard_indices(self, n: int) -> List[np.ndarray]:
+        """Contiguous split of range(n) across ranks (np.array_split
+        semantics, matching the reference's client→actor split,
+        simulator.py:223)."""
+        return np.array_split(np.arange(n), self.world_size)
+
+    def my_shard(self, items: Sequence) -> List:
+        idx = self.shard_indices(len(items))[self.rank]
+        return [items[i] for i in idx]
+
+    # ---------------------------------------------------------- collectives
+    def broadcast_flat(self, vec: torch.Tensor, src: int = 0) -> torch.Tensor:
+        if self.distributed:
+            dist.broadcast(vec, src=src)
+        return vec
+
+    def all_gather_rows(self, local: torch.Tensor, total_rows: int) -> torch.Tensor:
+        """Gather per-rank row blocks into the full [total_rows, d] matrix.
+
+        Rank r contributes the rows of shard r (contiguous).  Shards may be
+        uneven; blocks are padded to the max shard size for the collective
+        (NCCL all_gather needs equal shapes) and sliced after.
+        """
+        if not self.distributed:
+            return local
+        d = local.shape[1]
+        sizes = [len(s) for s in self.shard_indices(total_rows)]
+        kmax = max(sizes)
+        if local.shape[0] < kmax:
+            pad = torch.zeros(kmax - local.shape[0], d, device=local.device,
+                              dtype=local.dtype)
+            send = torch.cat([local, pad], dim=0).contiguous()
+        else:
+            send = local.contiguous()
+        out = torch.empty(self.world_size * kmax, d, device=local.device,
+                          dtype=local.dtype)
+        dist.all_gather_into_tensor(out, send)
+        blocks = [out[r * kmax: r * kmax + sizes[r]] for r in range(self.world_size)]
+        return torch.cat(blocks, dim=0)
+
+    def all_reduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
+        if self.distributed:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM if op == "sum" else dist.ReduceOp.MAX)
+        return t
+
+    def all_gather_object(self, obj) -> List:
+        if not self.distributed:
+            return [obj]
+        out = [None] * self.world_size
+        dist.all_gather_object(out, obj)
+        return out
+
+    def max_over_ranks(self, value: float) -> float:
+        if not self.distributed:
+            return value
+        t = torch.tensor([value], dtype=torch.float64)
+        # gloo handles CPU tensors; nccl needs device tensors
+        if dist.get_backend() == "nccl":
+            t = t.to(self.device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return float(t.item())
