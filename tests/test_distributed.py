@@ -1,0 +1,83 @@
+"""Multi-process distributed tests over gloo (CPU CI stand-in for the
+8-rank RCCL runtime — same code paths, world_size=2).
+
+Key invariant: sharding clients across ranks must not change results —
+every rank ends with the same θ, equal to the world_size=1 run.
+"""
+import os
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+from blades_amd.models import MLP
+
+WORLD = 2
+
+
+def _single_rank_theta(seed=5):
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    ds = SyntheticFLDataset(num_clients=6, samples_per_client=16, batch_size=8,
+                            shape=(1, 28, 28), num_classes=10, seed=0)
+    sim = Simulator(ds, num_byzantine=2, attack="alie",
+                    attack_kws={"num_clients": 6, "num_byzantine": 2},
+                    aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                    log_path="/tmp/bl_dist_single", seed=seed)
+    torch.manual_seed(seed)
+    sim.run(MLP(), global_rounds=3, validate_interval=0, client_lr=0.1)
+    return sim.server.flat_parameters()
+
+
+def _worker(rank, world_size, port, out_q, seed=5):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world_size),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    ds = SyntheticFLDataset(num_clients=6, samples_per_client=16, batch_size=8,
+                            shape=(1, 28, 28), num_classes=10, seed=0)
+    sim = Simulator(ds, num_byzantine=2, attack="alie",
+                    attack_kws={"num_clients": 6, "num_byzantine": 2},
+                    aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                    log_path=f"/tmp/bl_dist_r{rank}", seed=seed)
+    torch.manual_seed(seed)
+    sim.run(MLP(), global_rounds=3, validate_interval=0, client_lr=0.1)
+    loss, top1 = sim.test_actor(3, batch_size=16)
+    # send by value: shared-memory tensors die with the worker process
+    out_q.put((rank, sim.server.flat_parameters().numpy().copy(), loss, top1))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_world2_matches_single_rank():
+    theta1 = _single_rank_theta()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29611
+    procs = [ctx.Process(target=_worker, args=(r, WORLD, port, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(WORLD):
+        rank, theta, loss, top1 = q.get(timeout=240)
+        results[rank] = (theta, loss, top1)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    # all ranks converge to the same parameters
+    t0, l0, a0 = results[0]
+    t1, l1, a1 = results[1]
+    t0, t1 = torch.from_numpy(t0), torch.from_numpy(t1)
+    assert torch.equal(t0, t1)
+    assert l0 == pytest.approx(l1)
+    # and they match the single-rank run exactly (layout-invariant RNG)
+    assert torch.allclose(t0, theta1, atol=1e-6)

@@ -1,0 +1,67 @@
+"""Model zoo: architecture parity (param counts from BASELINE.md) and
+forward shapes."""
+import pytest
+import torch
+
+from blades_amd.models import (MLP, cct_2_3x2_32, get_model, num_params,
+                               resnet18, wide_resnet28_10)
+
+
+def test_mlp_param_count_exact():
+    # reference MLP is 784-64-128-10 → 59,850 params
+    assert num_params(MLP()) == 59850
+
+
+def test_resnet18_param_count():
+    n = num_params(resnet18())
+    assert abs(n - 11.2e6) / 11.2e6 < 0.01  # ≈11.2M
+
+
+def test_wrn28_10_param_count():
+    n = num_params(wide_resnet28_10())
+    assert abs(n - 36.5e6) / 36.5e6 < 0.01  # ≈36.5M
+
+
+def test_cct_param_count():
+    n = num_params(cct_2_3x2_32())
+    assert abs(n - 284e3) / 284e3 < 0.02  # ≈284k
+
+
+@pytest.mark.parametrize("name,shape,classes", [
+    ("mlp", (2, 1, 28, 28), 10),
+    ("resnet18", (2, 3, 32, 32), 10),
+    ("wrn28_10", (2, 3, 32, 32), 10),
+    ("cct", (2, 3, 32, 32), 10),
+])
+def test_forward_shapes(name, shape, classes):
+    m = get_model(name)
+    out = m(torch.randn(*shape))
+    assert out.shape == (shape[0], classes)
+
+
+def test_cct_sequence_length_is_64():
+    m = cct_2_3x2_32()
+    assert m.seq_len == 64  # 32x32 through two stride-2 pools → 8x8
+
+
+def test_norm_variants():
+    for norm in ["batch", "batch-local", "group"]:
+        m = resnet18(norm=norm)
+        out = m(torch.randn(2, 3, 32, 32))
+        assert out.shape == (2, 10)
+
+
+def test_num_classes_knob():
+    m = wide_resnet28_10(num_classes=100)
+    assert m(torch.randn(2, 3, 32, 32)).shape == (2, 100)
+
+
+def test_backward_runs():
+    for name in ["mlp", "resnet18", "cct"]:
+        m = get_model(name)
+        shape = (2, 1, 28, 28) if name == "mlp" else (2, 3, 32, 32)
+        loss = torch.nn.functional.cross_entropy(
+            m(torch.randn(*shape)), torch.tensor([1, 2]))
+        loss.backward()
+        assert all(p.grad is not None for p in m.parameters()
+                   if p.requires_grad)
