@@ -23,7 +23,7 @@ class SyntheticFLDataset:
     def __init__(self, num_clients: int = 100, samples_per_client: int = 64,
                  batch_size: int = 32, shape: Sequence[int] = (3, 32, 32),
                  num_classes: int = 10, seed: int = 0, device: str = "cpu",
-                 test_samples_per_client: int = 32):
+                 test_samples_per_client: int = 32, learnable: bool = False):
         self.num_clients = num_clients
         self.samples_per_client = samples_per_client
         self.batch_size = batch_size
@@ -32,6 +32,10 @@ class SyntheticFLDataset:
         self.seed = seed
         self.device = torch.device(device)
         self.test_samples_per_client = test_samples_per_client
+        # learnable=True labels samples with a fixed random linear teacher
+        # (same teacher for train and test), so eval loss can actually drop
+        self.learnable = learnable
+        self._teacher = None
         self._clients = list(range(num_clients))
         self._cursor = [0] * num_clients
         # lazy pools: client -> (X [S,*shape], y [S])
@@ -46,7 +50,17 @@ class SyntheticFLDataset:
         g = torch.Generator(device=self.device)
         g.manual_seed(client_philox_seed(self.seed, int(u_id), 0, tag=tag))
         X = torch.randn((n, *self.shape), generator=g, device=self.device)
-        y = torch.randint(0, self.num_classes, (n,), generator=g, device=self.device)
+        if self.learnable:
+            if self._teacher is None:
+                gt = torch.Generator(device=self.device)
+                gt.manual_seed(client_philox_seed(self.seed, 0, 0, tag=99))
+                dim = int(torch.tensor(self.shape).prod())
+                self._teacher = torch.randn((dim, self.num_classes),
+                                            generator=gt, device=self.device)
+            y = (X.flatten(1) @ self._teacher).argmax(dim=1)
+        else:
+            y = torch.randint(0, self.num_classes, (n,), generator=g,
+                              device=self.device)
         return X, y
 
     def _train_pool(self, u_id: int):
