@@ -1,0 +1,163 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — BASELINE.json headline metric.
+
+Metric: global rounds/sec on CIFAR10-shaped synthetic data, ResNet-18
+FedSGD, 100 clients per GPU (weak scaling) with 20% ALIE attackers and
+TrimmedMean aggregation (BASELINE.json config 2 at N=1; the same per-GPU
+work at N=2/4/8).
+
+Contract (driver): ``python bench.py --gpus N --steps K --warmup W``;
+for N>1 launched under torch.distributed.run with one rank per GPU.
+W untimed warmup rounds, then EXACTLY K timed rounds bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX time over ranks;
+rank 0 prints one JSON line.
+
+Compute dtype is fp32 — the reference trains in plain fp32 PyTorch, and
+gfx950 has no TF32; using bf16 would be below-reference precision and
+invalid per the bench rules.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--clients-per-gpu", type=int, default=100)
+    p.add_argument("--byz-frac", type=float, default=0.2)
+    p.add_argument("--batch", type=int, default=32)
+    p.add_argument("--local-steps", type=int, default=1)
+    p.add_argument("--model", type=str, default="resnet18")
+    p.add_argument("--aggregator", type=str, default="trimmedmean")
+    p.add_argument("--attack", type=str, default="alie")
+    p.add_argument("--client-chunk", type=int, default=None)
+    p.add_argument("--num-classes", type=int, default=10)
+    p.add_argument("--samples-per-client", type=int, default=64)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    n_gpus = max(args.gpus, world_size)
+    rank = int(os.environ.get("RANK", "0"))
+
+    use_cuda = torch.cuda.is_available()
+    device_kw = {}
+    if not use_cuda:
+        # CPU smoke mode (scaled down so it finishes in seconds)
+        args.clients_per_gpu = min(args.clients_per_gpu, 8)
+        args.model = "mlp"
+
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import get_model
+
+    total_clients = args.clients_per_gpu * n_gpus
+    num_byz = int(round(args.byz_frac * total_clients))
+
+    shape = (1, 28, 28) if args.model == "mlp" else (3, 32, 32)
+    model_kw = {"num_classes": args.num_classes}
+    if args.model in ("resnet18", "wrn28_10"):
+        model_kw["norm"] = "batch-local"
+
+    agg_kws = {}
+    if args.aggregator == "trimmedmean":
+        agg_kws = {"nb": max(1, num_byz)}
+    elif args.aggregator in ("krum", "multikrum"):
+        agg_kws = {"num_clients": total_clients, "num_byzantine": num_byz}
+    attack_kws = {}
+    if args.attack == "alie":
+        attack_kws = {"num_clients": total_clients, "num_byzantine": num_byz}
+
+    sim = Simulator(
+        dataset=SyntheticFLDataset(
+            num_clients=total_clients,
+            samples_per_client=args.samples_per_client,
+            batch_size=args.batch, shape=shape,
+            num_classes=args.num_classes, seed=0, device="cpu"),
+        num_byzantine=num_byz,
+        attack=args.attack,
+        attack_kws=attack_kws,
+        aggregator=args.aggregator,
+        aggregator_kws=agg_kws,
+        use_cuda=use_cuda,
+        log_path=os.environ.get("BLADES_BENCH_LOG", "/tmp/blades_bench_logs"),
+        seed=1234,
+        client_chunk=args.client_chunk,
+    )
+    device = sim.device
+    # move the synthetic pools to the device and pre-materialize the shard
+    sim.dataset.device = device
+    shard_ids = [c.id() for c in sim.runtime.my_shard(sim.get_clients())]
+    sim.dataset.materialize(shard_ids)
+
+    model = get_model(args.model, **model_kw)
+    # build engines/server without running any round
+    sim.run(model, global_rounds=0, local_steps=args.local_steps,
+            validate_interval=0, client_lr=0.1, server_lr=1.0)
+
+    clients = sim.get_clients()
+
+    def one_round(r):
+        sim.train_round(r, args.local_steps, clients, 0.1)
+
+    rt = sim.runtime
+    for r in range(args.warmup):
+        one_round(r + 1)
+
+    rt.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(device)
+    t0 = time.perf_counter()
+    for r in range(args.steps):
+        one_round(args.warmup + r + 1)
+    rt.barrier()
+    if use_cuda:
+        torch.cuda.synchronize(device)
+    elapsed = time.perf_counter() - t0
+
+    elapsed = rt.max_over_ranks(elapsed)
+    rounds_per_sec = args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": f"global rounds/sec ({args.model} FedSGD, "
+                      f"{args.clients_per_gpu} clients/GPU, "
+                      f"{args.attack}+{args.aggregator})",
+            "value": rounds_per_sec,
+            "unit": "rounds/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * args.clients_per_gpu * n_gpus,
+                "clients": args.clients_per_gpu * n_gpus,
+                "byzantine": num_byz,
+                "attack": args.attack,
+                "aggregator": args.aggregator,
+                "local_steps": args.local_steps,
+                "seq_len": None,
+                "parallelism": f"client-sharded dp{n_gpus}",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -58,6 +58,12 @@ def _force_torch() -> bool:
     return os.environ.get("BLADES_AMD_FORCE_TORCH", "0") == "1"
 
 
+def _prep(U: Tensor) -> Tensor:
+    """Kernels accept any row stride with contiguous rows (padded slabs pass
+    through zero-copy); only fully transposed/odd layouts get materialized."""
+    return U if U.stride(1) == 1 else U.contiguous()
+
+
 def _route(U: Tensor):
     """Return the HIP extension if this tensor must run on it, else None."""
     if U.is_cuda and not _force_torch():
@@ -79,14 +85,14 @@ def _route(U: Tensor):
 def col_mean(U: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.col_mean(U.contiguous())
+        return ext.col_mean(_prep(U))
     return _ref.col_mean(U)
 
 
 def col_median(U: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.col_median(U.contiguous())
+        return ext.col_median(_prep(U))
     return _ref.col_median(U)
 
 
@@ -96,21 +102,21 @@ def trimmed_mean(U: Tensor, b: int) -> Tensor:
         raise ValueError(f"trimmed_mean needs K > 2b (K={K}, b={b})")
     ext = _route(U)
     if ext is not None:
-        return ext.trimmed_mean(U.contiguous(), b)
+        return ext.trimmed_mean(_prep(U), b)
     return _ref.trimmed_mean(U, b)
 
 
 def weighted_col_sum(U: Tensor, w: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.weighted_col_sum(U.contiguous(), w.to(U).contiguous())
+        return ext.weighted_col_sum(_prep(U), w.to(U).contiguous())
     return _ref.weighted_col_sum(U, w)
 
 
 def masked_col_mean(U: Tensor, mask: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.masked_col_mean(U.contiguous(), mask.to(torch.bool).contiguous())
+        return ext.masked_col_mean(_prep(U), mask.to(torch.bool).contiguous())
     return _ref.masked_col_mean(U, mask)
 
 
@@ -118,7 +124,7 @@ def masked_col_mean_std(U: Tensor, mask: Tensor,
                         unbiased: bool = True) -> Tuple[Tensor, Tensor]:
     ext = _route(U)
     if ext is not None:
-        return ext.masked_col_mean_std(U.contiguous(),
+        return ext.masked_col_mean_std(_prep(U),
                                        mask.to(torch.bool).contiguous(), unbiased)
     return _ref.masked_col_mean_std(U, mask, unbiased)
 
@@ -128,7 +134,7 @@ def masked_col_mean_std(U: Tensor, mask: Tensor,
 def row_sq_norms(U: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.row_sq_norms(U.contiguous())
+        return ext.row_sq_norms(_prep(U))
     return _ref.row_sq_norms(U)
 
 
@@ -139,21 +145,21 @@ def row_norms(U: Tensor) -> Tensor:
 def row_diff_norms(U: Tensor, z: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.row_diff_sq_norms(U.contiguous(), z.contiguous()).sqrt()
+        return ext.row_diff_sq_norms(_prep(U), z.contiguous()).sqrt()
     return _ref.row_diff_norms(U, z)
 
 
 def row_dots(U: Tensor, v: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.row_dots(U.contiguous(), v.contiguous())
+        return ext.row_dots(_prep(U), v.contiguous())
     return _ref.row_dots(U, v)
 
 
 def pairwise_sq_dists(U: Tensor) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        G = ext.gram(U.contiguous())
+        G = ext.gram(_prep(U))
         sq = G.diagonal()
         D = sq.unsqueeze(0) + sq.unsqueeze(1) - 2 * G
         D.fill_diagonal_(0)
@@ -165,15 +171,15 @@ def gram(U: Tensor) -> Tensor:
     """U @ U.T — MFMA f32 on GPU."""
     ext = _route(U)
     if ext is not None:
-        return ext.gram(U.contiguous())
+        return ext.gram(_prep(U))
     return U @ U.t()
 
 
 def cos_sim_gram(U: Tensor, eps: float = 1e-8) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        norms = ext.row_sq_norms(U.contiguous()).sqrt().clamp_min(eps)
-        G = ext.gram(U.contiguous())
+        norms = ext.row_sq_norms(_prep(U)).sqrt().clamp_min(eps)
+        G = ext.gram(_prep(U))
         G = G / norms.unsqueeze(0) / norms.unsqueeze(1)
         return G.clamp_(-1.0, 1.0)
     return _ref.cos_sim_gram(U, eps)
@@ -182,7 +188,14 @@ def cos_sim_gram(U: Tensor, eps: float = 1e-8) -> Tensor:
 def centered_clip_iter(U: Tensor, v: Tensor, tau: float) -> Tensor:
     ext = _route(U)
     if ext is not None:
-        return ext.centered_clip_iter(U.contiguous(), v.contiguous(), float(tau))
+        # composition of HIP primitives (K7): one row-norm pass + one
+        # weighted column pass; v' = v·(1 − Σs/K) + Σ (s_k/K)·u_k with
+        # s_k = min(1, τ/‖u_k − v‖)
+        K = U.shape[0]
+        vc = v.contiguous()
+        norms = ext.row_diff_sq_norms(_prep(U), vc).sqrt().clamp_min(1e-12)
+        scale = torch.clamp(tau / norms, max=1.0) / K
+        return vc * (1.0 - scale.sum()) + ext.weighted_col_sum(U, scale)
     return _ref.centered_clip_iter(U, v, tau)
 
 

@@ -1,0 +1,556 @@
+// blades_amd CDNA4 (gfx950 / MI355X) kernel library.
+//
+// Hand-written HIP kernels for the robust-aggregation / attack hot path
+// (SURVEY.md §2.4 K1-K11).  All kernels operate on the stacked update
+// matrix U ∈ R^{K×d} (fp32, row stride ld >= d), which the runtime keeps
+// resident in HBM3E.  Design notes per kernel at the definition.
+//
+// Conventions:
+//  * wavefront = 64 (CDNA), block sizes are multiples of 64
+//  * memory-bound kernels: float4 loads (16 B/lane) over the vectorizable
+//    column span, scalar tail; grid capped ~2048 blocks + grid-stride
+//  * the Gram kernel uses the f32 MFMA (v_mfma_f32_32x32x2_f32) — exact
+//    fp32 numerics at the 157 TF f32 rate, LDS-staged tiles
+//
+// Build: hipcc --offload-arch=gfx950 via torch.utils.cpp_extension (see
+// setup.py).  No CUDA shims, no hipify — this file is native HIP.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <tuple>
+
+#define CHECK_IN(x)                                                     \
+  TORCH_CHECK((x).is_cuda(), #x " must be on GPU");                     \
+  TORCH_CHECK((x).scalar_type() == at::kFloat, #x " must be fp32");
+
+namespace {
+
+constexpr int kMaxBlocks = 4096;  // 256 CU x ~16 blocks
+
+__host__ __device__ inline long long cdiv(long long a, long long b) {
+  return (a + b - 1) / b;
+}
+
+// ---------------------------------------------------------------------------
+// Column reductions (K1, GeoMed/FLTrust accumulation, ALIE/IPM stats)
+// ---------------------------------------------------------------------------
+// One thread owns one (vector of 4) column(s); lane i and lane i+1 touch
+// adjacent float4s -> fully coalesced 16 B/lane streams.  The K-loop walks
+// rows at stride ld.  MODE selects the accumulation:
+//   0: mean            out0[j] = (1/K) sum_k U[k][j]
+//   1: weighted sum    out0[j] = sum_k w[k] U[k][j]
+//   2: masked mean     out0[j] = (1/|M|) sum_{k in M} U[k][j]
+//   3: masked mean+var out0=mean, out1=std over mask (Welford-free two-sum:
+//      fp32 sum + sumsq is adequate at K<=1e4 given |U| ~ lr*grad scales;
+//      matches torch.std to ~1e-6 in tests)
+
+template <int MODE, bool VEC>
+__global__ void col_reduce_kernel(const float* __restrict__ U,
+                                  const float* __restrict__ w,
+                                  const bool* __restrict__ mask,
+                                  float* __restrict__ out0,
+                                  float* __restrict__ out1,
+                                  long long K, long long d, long long ld,
+                                  float inv_count, bool unbiased,
+                                  float count) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+
+  if (VEC) {
+    const long long d4 = d / 4;
+    for (long long j4 = tid; j4 < d4; j4 += nthreads) {
+      float4 acc = make_float4(0.f, 0.f, 0.f, 0.f);
+      float4 acc2 = make_float4(0.f, 0.f, 0.f, 0.f);
+      const float* col = U + 4 * j4;
+      for (long long k = 0; k < K; ++k) {
+        if (MODE == 2 || MODE == 3) {
+          if (!mask[k]) continue;
+        }
+        const float4 v = *reinterpret_cast<const float4*>(col + k * ld);
+        const float wk = (MODE == 1) ? w[k] : 1.f;
+        acc.x += wk * v.x; acc.y += wk * v.y;
+        acc.z += wk * v.z; acc.w += wk * v.w;
+        if (MODE == 3) {
+          acc2.x += v.x * v.x; acc2.y += v.y * v.y;
+          acc2.z += v.z * v.z; acc2.w += v.w * v.w;
+        }
+      }
+      if (MODE == 0 || MODE == 2) {
+        acc.x *= inv_count; acc.y *= inv_count;
+        acc.z *= inv_count; acc.w *= inv_count;
+      }
+      *reinterpret_cast<float4*>(out0 + 4 * j4) = acc;
+      if (MODE == 3) {
+        const float denom = unbiased ? (count - 1.f) : count;
+        float4 sd;
+        sd.x = sqrtf(fmaxf((acc2.x - acc.x * acc.x / count), 0.f) / denom);
+        sd.y = sqrtf(fmaxf((acc2.y - acc.y * acc.y / count), 0.f) / denom);
+        sd.z = sqrtf(fmaxf((acc2.z - acc.z * acc.z / count), 0.f) / denom);
+        sd.w = sqrtf(fmaxf((acc2.w - acc.w * acc.w / count), 0.f) / denom);
+        *reinterpret_cast<float4*>(out1 + 4 * j4) = sd;
+        // mean output for MODE 3
+        float4 mu;
+        mu.x = acc.x / count; mu.y = acc.y / count;
+        mu.z = acc.z / count; mu.w = acc.w / count;
+        *reinterpret_cast<float4*>(out0 + 4 * j4) = mu;
+      }
+    }
+  }
+
+  // scalar span: the whole matrix when !VEC, else just the tail columns
+  const long long j_begin = VEC ? (d / 4) * 4 : 0;
+  for (long long j = j_begin + tid; j < d; j += nthreads) {
+    float acc = 0.f, acc2 = 0.f;
+    for (long long k = 0; k < K; ++k) {
+      if (MODE == 2 || MODE == 3) {
+        if (!mask[k]) continue;
+      }
+      const float v = U[k * ld + j];
+      const float wk = (MODE == 1) ? w[k] : 1.f;
+      acc += wk * v;
+      if (MODE == 3) acc2 += v * v;
+    }
+    if (MODE == 0 || MODE == 2) acc *= inv_count;
+    if (MODE == 3) {
+      const float denom = unbiased ? (count - 1.f) : count;
+      out1[j] = sqrtf(fmaxf((acc2 - acc * acc / count), 0.f) / denom);
+      out0[j] = acc / count;
+    } else {
+      out0[j] = acc;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K3/K2 — coordinate-wise trimmed sum / median selection
+// ---------------------------------------------------------------------------
+// One thread owns one coordinate and streams its column (coalesced across
+// lanes), maintaining the b smallest and b largest values seen so far in
+// per-thread LDS scratch.  Expected replacement work is b·ln(K/b) per
+// column, so the kernel stays memory-bound for the target shapes
+// (K=100, b=20 headline config).  Output: trimmed mean over the middle
+// K-2b values.  Median = b=(K-1)/2 (averages both middles for even K,
+// matching reference semantics, aggregators/median.py:23-25).
+//
+// LDS layout: per-thread strided (lo[i*BS + tid]) so each 32-lane group
+// hits 32 distinct banks (b32 banking).  NaNs must be sanitized upstream
+// (get_update applies nan_to_num — client.py:198 semantics).
+
+__global__ void trimmed_select_kernel(const float* __restrict__ U,
+                                      float* __restrict__ out,
+                                      long long K, long long d, long long ld,
+                                      int b) {
+  extern __shared__ float smem[];
+  const int BS = blockDim.x;
+  float* lo = smem + threadIdx.x;            // b floats, stride BS
+  float* hi = smem + (size_t)b * BS + threadIdx.x;
+
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+  const float inv = 1.f / (float)(K - 2 * b);
+
+  for (long long j = tid; j < d; j += nthreads) {
+    float sum = 0.f;
+    int nlo = 0, nhi = 0;
+    float lo_max = -INFINITY, hi_min = INFINITY;
+    int lo_max_i = 0, hi_min_i = 0;
+
+    for (long long k = 0; k < K; ++k) {
+      const float v = U[k * ld + j];
+      sum += v;
+      if (b > 0) {
+        // b smallest
+        if (nlo < b) {
+          lo[nlo * BS] = v;
+          if (v > lo_max) { lo_max = v; lo_max_i = nlo; }
+          ++nlo;
+        } else if (v < lo_max) {
+          lo[lo_max_i * BS] = v;
+          lo_max = lo[0];  lo_max_i = 0;
+          for (int i = 1; i < b; ++i) {
+            const float x = lo[i * BS];
+            if (x > lo_max) { lo_max = x; lo_max_i = i; }
+          }
+        }
+        // b largest
+        if (nhi < b) {
+          hi[nhi * BS] = v;
+          if (v < hi_min) { hi_min = v; hi_min_i = nhi; }
+          ++nhi;
+        } else if (v > hi_min) {
+          hi[hi_min_i * BS] = v;
+          hi_min = hi[0];  hi_min_i = 0;
+          for (int i = 1; i < b; ++i) {
+            const float x = hi[i * BS];
+            if (x < hi_min) { hi_min = x; hi_min_i = i; }
+          }
+        }
+      }
+    }
+    float trim = 0.f;
+    for (int i = 0; i < b; ++i) trim += lo[i * BS] + hi[i * BS];
+    out[j] = (sum - trim) * inv;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Row reductions (K6 Weiszfeld distances, K7 clip norms, K9 FLTrust dots)
+// ---------------------------------------------------------------------------
+// grid = (splits, K): block (s, k) reduces slice s of row k with float4
+// loads + wave shuffle + one atomicAdd per block (out zero-initialized).
+//   MODE 0: sum U[k][j]^2        MODE 1: sum (U[k][j]-z[j])^2
+//   MODE 2: sum U[k][j]*v[j]
+
+template <int MODE, bool VEC>
+__global__ void row_reduce_kernel(const float* __restrict__ U,
+                                  const float* __restrict__ z,
+                                  float* __restrict__ out,
+                                  long long K, long long d, long long ld) {
+  const long long k = blockIdx.y;
+  const float* row = U + k * ld;
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+
+  float acc = 0.f;
+  if (VEC) {
+    const long long d4 = d / 4;
+    for (long long j4 = tid; j4 < d4; j4 += nthreads) {
+      const float4 v = *reinterpret_cast<const float4*>(row + 4 * j4);
+      if (MODE == 0) {
+        acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+      } else {
+        const float4 u = *reinterpret_cast<const float4*>(z + 4 * j4);
+        if (MODE == 1) {
+          const float a = v.x - u.x, bq = v.y - u.y, c = v.z - u.z, e = v.w - u.w;
+          acc += a * a + bq * bq + c * c + e * e;
+        } else {
+          acc += v.x * u.x + v.y * u.y + v.z * u.z + v.w * u.w;
+        }
+      }
+    }
+  }
+  const long long j_begin = VEC ? (d / 4) * 4 : 0;
+  for (long long j = j_begin + tid; j < d; j += nthreads) {
+    const float v = row[j];
+    if (MODE == 0) acc += v * v;
+    else if (MODE == 1) { const float a = v - z[j]; acc += a * a; }
+    else acc += v * z[j];
+  }
+
+  // wave reduce (64-wide), then LDS cross-wave, then one atomic per block
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  __shared__ float wsum[16];
+  const int wid = threadIdx.x / 64;
+  if ((threadIdx.x & 63) == 0) wsum[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int i = 0; i < (int)(blockDim.x / 64); ++i) s += wsum[i];
+    atomicAdd(&out[k], s);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K4/K8 — Gram matrix G = U U^T via f32 MFMA (v_mfma_f32_32x32x2_f32)
+// ---------------------------------------------------------------------------
+// grid = (tile pairs (it,jt), d-slices).  Block = 256 threads (4 waves).
+// Per 32-column chunk of the slice: stage the two 32-row tiles in LDS
+// (coalesced float loads, rows padded to 33 floats against bank conflicts),
+// then each wave runs 4 of the 16 k-steps (2 contraction columns per MFMA)
+// on its own accumulator; wave partials combine through LDS and one
+// atomicAdd per output element into G (fp32 atomics: bit-level run-to-run
+// variation ~1e-7 relative — documented, tested with tolerance).
+// Numerics: MFMA f32 is an exact fmaf chain (guide §3), same class as a
+// VALU dot product.
+
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+constexpr int GRAM_BD = 32;  // contraction chunk (columns per LDS stage)
+
+__global__ void gram_mfma_kernel(const float* __restrict__ U,
+                                 float* __restrict__ G,
+                                 long long K, long long d, long long ld,
+                                 int ntiles, long long slice_len) {
+  __shared__ float As[32][33];
+  __shared__ float Bs[32][33];
+  __shared__ float Gs[32][33];
+
+  const int pair = blockIdx.x;
+  const int it = pair / ntiles;
+  const int jt = pair % ntiles;
+  if (jt < it) return;  // symmetric: compute upper triangle, mirror on host
+
+  const long long c_begin = (long long)blockIdx.y * slice_len;
+  const long long c_end = (c_begin + slice_len < d) ? c_begin + slice_len : d;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+
+  f32x16 acc = {};
+
+  for (long long c0 = c_begin; c0 < c_end; c0 += GRAM_BD) {
+    const int cols = (int)min((long long)GRAM_BD, c_end - c0);
+    // stage tiles: 256 threads load 32x32 each (4 elements/thread)
+    __syncthreads();
+    for (int e = threadIdx.x; e < 32 * GRAM_BD; e += blockDim.x) {
+      const int r = e / GRAM_BD, cc = e % GRAM_BD;
+      const long long gi = (long long)it * 32 + r;
+      const long long gj = (long long)jt * 32 + r;
+      As[r][cc] = (gi < K && cc < cols) ? U[gi * ld + c0 + cc] : 0.f;
+      Bs[r][cc] = (gj < K && cc < cols) ? U[gj * ld + c0 + cc] : 0.f;
+    }
+    __syncthreads();
+    // 16 k-steps of 2 columns; wave w takes steps w, w+4, w+8, w+12
+    const int row = lane & 31;
+    const int ksel = lane >> 5;  // 0/1: which of the 2 contraction columns
+    for (int s = wid; s < GRAM_BD / 2; s += 4) {
+      const int c = 2 * s + ksel;
+      const float a = As[row][c];
+      const float b = Bs[row][c];
+      acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+    }
+  }
+
+  // combine the 4 wave partials in LDS, then one atomicAdd per element
+  __syncthreads();
+  for (int e = threadIdx.x; e < 32 * 33; e += blockDim.x)
+    (&Gs[0][0])[e] = 0.f;
+  __syncthreads();
+  {
+    const int col = lane & 31;
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      atomicAdd(&Gs[row][col], acc[r]);
+    }
+  }
+  __syncthreads();
+  for (int e = threadIdx.x; e < 32 * 32; e += blockDim.x) {
+    const int r = e / 32, c = e % 32;
+    const long long gi = (long long)it * 32 + r;
+    const long long gj = (long long)jt * 32 + c;
+    if (gi < K && gj < K) {
+      atomicAdd(&G[gi * K + gj], Gs[r][c]);
+      if (it != jt)  // mirror off-diagonal tiles (diagonal tiles are full)
+        atomicAdd(&G[gj * K + gi], Gs[r][c]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+struct UView {
+  const float* ptr;
+  long long K, d, ld;
+  bool vec;  // float4 path legal (16B-aligned rows)
+};
+
+UView view_of(const torch::Tensor& U) {
+  CHECK_IN(U);
+  TORCH_CHECK(U.dim() == 2, "U must be 2-D");
+  TORCH_CHECK(U.stride(1) == 1, "U rows must be innermost-contiguous");
+  UView v;
+  v.ptr = U.data_ptr<float>();
+  v.K = U.size(0);
+  v.d = U.size(1);
+  v.ld = U.stride(0);
+  v.vec = (v.ld % 4 == 0) &&
+          (reinterpret_cast<uintptr_t>(v.ptr) % 16 == 0);
+  return v;
+}
+
+static inline int col_grid(long long d, int block) {
+  return (int)std::min<long long>(cdiv(d, block), kMaxBlocks);
+}
+
+torch::Tensor col_mean(torch::Tensor U) {
+  auto v = view_of(U);
+  auto out = torch::empty({v.d}, U.options());
+  const int BS = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const float invK = 1.f / (float)v.K;
+  if (v.vec)
+    col_reduce_kernel<0, true><<<col_grid(v.d / 4 + 1, BS), BS, 0, stream>>>(
+        v.ptr, nullptr, nullptr, out.data_ptr<float>(), nullptr,
+        v.K, v.d, v.ld, invK, false, (float)v.K);
+  else
+    col_reduce_kernel<0, false><<<col_grid(v.d, BS), BS, 0, stream>>>(
+        v.ptr, nullptr, nullptr, out.data_ptr<float>(), nullptr,
+        v.K, v.d, v.ld, invK, false, (float)v.K);
+  return out;
+}
+
+torch::Tensor weighted_col_sum(torch::Tensor U, torch::Tensor w) {
+  auto v = view_of(U);
+  CHECK_IN(w);
+  TORCH_CHECK(w.numel() == v.K, "weight length mismatch");
+  auto wc = w.contiguous();
+  auto out = torch::empty({v.d}, U.options());
+  const int BS = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (v.vec)
+    col_reduce_kernel<1, true><<<col_grid(v.d / 4 + 1, BS), BS, 0, stream>>>(
+        v.ptr, wc.data_ptr<float>(), nullptr, out.data_ptr<float>(), nullptr,
+        v.K, v.d, v.ld, 1.f, false, (float)v.K);
+  else
+    col_reduce_kernel<1, false><<<col_grid(v.d, BS), BS, 0, stream>>>(
+        v.ptr, wc.data_ptr<float>(), nullptr, out.data_ptr<float>(), nullptr,
+        v.K, v.d, v.ld, 1.f, false, (float)v.K);
+  return out;
+}
+
+torch::Tensor masked_col_mean(torch::Tensor U, torch::Tensor mask) {
+  auto v = view_of(U);
+  TORCH_CHECK(mask.scalar_type() == at::kBool && mask.numel() == v.K);
+  auto mc = mask.contiguous();
+  const float cnt = (float)mask.sum().item<long>();
+  auto out = torch::empty({v.d}, U.options());
+  const int BS = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const float inv = cnt > 0 ? 1.f / cnt : 0.f;
+  if (v.vec)
+    col_reduce_kernel<2, true><<<col_grid(v.d / 4 + 1, BS), BS, 0, stream>>>(
+        v.ptr, nullptr, mc.data_ptr<bool>(), out.data_ptr<float>(), nullptr,
+        v.K, v.d, v.ld, inv, false, cnt);
+  else
+    col_reduce_kernel<2, false><<<col_grid(v.d, BS), BS, 0, stream>>>(
+        v.ptr, nullptr, mc.data_ptr<bool>(), out.data_ptr<float>(), nullptr,
+        v.K, v.d, v.ld, inv, false, cnt);
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor>
+masked_col_mean_std(torch::Tensor U, torch::Tensor mask, bool unbiased) {
+  auto v = view_of(U);
+  TORCH_CHECK(mask.scalar_type() == at::kBool && mask.numel() == v.K);
+  auto mc = mask.contiguous();
+  const float cnt = (float)mask.sum().item<long>();
+  TORCH_CHECK(cnt >= 2, "need >=2 masked rows for std");
+  auto mu = torch::empty({v.d}, U.options());
+  auto sd = torch::empty({v.d}, U.options());
+  const int BS = 256;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (v.vec)
+    col_reduce_kernel<3, true><<<col_grid(v.d / 4 + 1, BS), BS, 0, stream>>>(
+        v.ptr, nullptr, mc.data_ptr<bool>(), mu.data_ptr<float>(),
+        sd.data_ptr<float>(), v.K, v.d, v.ld, 1.f / cnt, unbiased, cnt);
+  else
+    col_reduce_kernel<3, false><<<col_grid(v.d, BS), BS, 0, stream>>>(
+        v.ptr, nullptr, mc.data_ptr<bool>(), mu.data_ptr<float>(),
+        sd.data_ptr<float>(), v.K, v.d, v.ld, 1.f / cnt, unbiased, cnt);
+  return {mu, sd};
+}
+
+torch::Tensor trimmed_mean_select(torch::Tensor U, long b) {
+  auto v = view_of(U);
+  TORCH_CHECK(v.K - 2 * b >= 1, "trimmed_mean needs K > 2b");
+  TORCH_CHECK(b >= 0, "b must be >= 0");
+  auto out = torch::empty({v.d}, U.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  // pick block size so the 2*b per-thread LDS floats fit in 64 KiB
+  int BS = 256;
+  size_t need = 2 * (size_t)b * 4 * BS;
+  while (BS > 64 && need > 64 * 1024) { BS /= 2; need /= 2; }
+  TORCH_CHECK(need <= 64 * 1024,
+              "b too large for the LDS selection kernel (use torch fallback)");
+  const int grid = col_grid(v.d, BS);
+  trimmed_select_kernel<<<grid, BS, need, stream>>>(
+      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b);
+  return out;
+}
+
+torch::Tensor trimmed_mean(torch::Tensor U, long b) {
+  return trimmed_mean_select(U, b);
+}
+
+torch::Tensor col_median(torch::Tensor U) {
+  auto v = view_of(U);
+  const long b = (v.K - 1) / 2;  // K-2b = 1 (odd K) or 2 (even K: avg both)
+  return trimmed_mean_select(U, b);
+}
+
+static torch::Tensor row_reduce(torch::Tensor U, const float* z, int mode) {
+  auto v = view_of(U);
+  auto out = torch::zeros({v.K}, U.options());
+  const int BS = 256;
+  // enough blocks to fill the chip even at small K
+  const int splits = (int)std::max<long long>(
+      1, std::min<long long>(cdiv(v.d, 4 * BS), cdiv(2048, v.K)));
+  dim3 grid(splits, (unsigned)v.K);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  if (v.vec) {
+    if (mode == 0)
+      row_reduce_kernel<0, true><<<grid, BS, 0, stream>>>(v.ptr, z, out.data_ptr<float>(), v.K, v.d, v.ld);
+    else if (mode == 1)
+      row_reduce_kernel<1, true><<<grid, BS, 0, stream>>>(v.ptr, z, out.data_ptr<float>(), v.K, v.d, v.ld);
+    else
+      row_reduce_kernel<2, true><<<grid, BS, 0, stream>>>(v.ptr, z, out.data_ptr<float>(), v.K, v.d, v.ld);
+  } else {
+    if (mode == 0)
+      row_reduce_kernel<0, false><<<grid, BS, 0, stream>>>(v.ptr, z, out.data_ptr<float>(), v.K, v.d, v.ld);
+    else if (mode == 1)
+      row_reduce_kernel<1, false><<<grid, BS, 0, stream>>>(v.ptr, z, out.data_ptr<float>(), v.K, v.d, v.ld);
+    else
+      row_reduce_kernel<2, false><<<grid, BS, 0, stream>>>(v.ptr, z, out.data_ptr<float>(), v.K, v.d, v.ld);
+  }
+  return out;
+}
+
+torch::Tensor row_sq_norms(torch::Tensor U) {
+  return row_reduce(U, nullptr, 0);
+}
+
+torch::Tensor row_diff_sq_norms(torch::Tensor U, torch::Tensor z) {
+  CHECK_IN(z);
+  auto zc = z.contiguous();
+  TORCH_CHECK(zc.numel() == U.size(1), "z length mismatch");
+  return row_reduce(U, zc.data_ptr<float>(), 1);
+}
+
+torch::Tensor row_dots(torch::Tensor U, torch::Tensor vv) {
+  CHECK_IN(vv);
+  auto vc = vv.contiguous();
+  TORCH_CHECK(vc.numel() == U.size(1), "v length mismatch");
+  return row_reduce(U, vc.data_ptr<float>(), 2);
+}
+
+torch::Tensor gram(torch::Tensor U) {
+  auto v = view_of(U);
+  auto G = torch::zeros({v.K, v.K}, U.options());
+  const int ntiles = (int)cdiv(v.K, 32);
+  const int npairs = ntiles * ntiles;  // lower triangle returns early
+  // slice d so total blocks ~ >=1024 for occupancy
+  long long nsl = std::max<long long>(1, 1024 / std::max(1, npairs));
+  long long slice = std::max<long long>(GRAM_BD, cdiv(v.d, nsl));
+  slice = cdiv(slice, GRAM_BD) * GRAM_BD;
+  nsl = cdiv(v.d, slice);
+  dim3 grid((unsigned)npairs, (unsigned)nsl);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  gram_mfma_kernel<<<grid, 256, 0, stream>>>(
+      v.ptr, G.data_ptr<float>(), v.K, v.d, v.ld, ntiles, slice);
+  return G;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "blades_amd CDNA4 HIP kernels (gfx950)";
+  m.def("col_mean", &col_mean, "column mean (K1)");
+  m.def("weighted_col_sum", &weighted_col_sum, "weighted column sum");
+  m.def("masked_col_mean", &masked_col_mean, "masked column mean (K11)");
+  m.def("masked_col_mean_std", &masked_col_mean_std,
+        "masked column mean+std (K10)");
+  m.def("trimmed_mean", &trimmed_mean, "coordinate-wise trimmed mean (K3)");
+  m.def("col_median", &col_median, "coordinate-wise median (K2)");
+  m.def("row_sq_norms", &row_sq_norms, "per-row squared norms");
+  m.def("row_diff_sq_norms", &row_diff_sq_norms,
+        "per-row squared distance to z (K6)");
+  m.def("row_dots", &row_dots, "per-row dot with v (K9)");
+  m.def("gram", &gram, "U U^T via f32 MFMA (K4/K8)");
+}
