@@ -73,12 +73,16 @@ class ParamSpec:
 
     # ------------------------------------------------------- batched slab
     def batched_views(self, slab: Tensor) -> Dict[str, Tensor]:
-        """Per-parameter [C, *shape] views into a contiguous [C, d] slab."""
-        assert slab.dim() == 2 and slab.shape[1] == self.d and slab.is_contiguous()
+        """Per-parameter [C, *shape] views into a [C, d] slab.
+
+        The slab may have row stride > d (padded slabs for 16B-aligned HIP
+        kernel access) as long as rows are innermost-contiguous."""
+        assert slab.dim() == 2 and slab.shape[1] == self.d and slab.stride(1) == 1
         C = slab.shape[0]
+        row_stride = slab.stride(0)
         out: Dict[str, Tensor] = {}
         for name, off, n, shape in zip(self.names, self.offsets, self.numels, self.shapes):
-            stride = (self.d,) + tuple(torch.empty(shape).stride())
+            stride = (row_stride,) + tuple(torch.empty(shape).stride())
             out[name] = slab.as_strided(size=(C, *shape), stride=stride,
                                         storage_offset=slab.storage_offset() + off)
         return out

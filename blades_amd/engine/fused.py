@@ -128,10 +128,13 @@ class FusedEngine:
         return X, Y
 
     def run_round(self, theta: Tensor, clients: List[BladesClient], dataset,
-                  local_steps: int, lr: float) -> Tensor:
+                  local_steps: int, lr: float,
+                  out: Optional[Tensor] = None) -> Tensor:
         """Returns the update slab U = θ_after − θ [C, d] on ``self.device``.
 
         ``theta`` is the flat global parameter vector (device-resident).
+        ``out``: optional [C, d] destination (may be a padded-row view) the
+        updates are written into directly — saves a slab copy per round.
         """
         C = len(clients)
         clamp_hi, grad_sign, target_tfms = self._client_vectors(clients)
@@ -165,7 +168,8 @@ class FusedEngine:
             X, Y = steps_data[0]
             Y = fix_targets(Y)
             grads = grad_fn(params, X, Y, clamp_hi, grad_sign)
-            U = torch.empty((C, self.spec.d), device=self.device)
+            U = out if out is not None else torch.empty(
+                (C, self.spec.d), device=self.device)
             views = self.spec.batched_views(U)
             with torch.no_grad():
                 for name, g in zip(self.spec.names, grads):
@@ -186,6 +190,9 @@ class FusedEngine:
                 torch._foreach_add_([slab_views[n] for n in self.spec.names],
                                     list(grads), alpha=-lr)
         slab.sub_(theta.unsqueeze(0))
+        if out is not None:
+            out.copy_(slab)
+            return out
         return slab
 
     # ------------------------------------------------------------------ eval

@@ -257,28 +257,38 @@ class Simulator:
         rows = {c.id(): i for i, c in enumerate(all_clients)}
         shard = rt.my_shard(all_clients)
 
+        # slab rows padded to a float4 multiple: the HIP kernels take the
+        # 16B-vectorized path on every row while views stay zero-copy
+        d = self._spec.d
+        d_pad = -(-d // 4) * 4
         with trace_range("blades/local_train"):
             theta = self.server.flat_parameters(device=self.device)
             fusable, custom = split_fusable(shard)
             if self._engine_choice == "loop":
                 fusable, custom = [], shard
-            U_local = torch.empty(len(shard), self._spec.d, device=self.device)
+            buf_local = torch.zeros(len(shard), d_pad, device=self.device)
+            U_local = buf_local[:, :d]
             local_pos = {c.id(): i for i, c in enumerate(shard)}
-            if fusable:
-                Uf = self._fused.run_round(theta, fusable, self.dataset,
-                                           local_steps, lr)
-                idx = torch.tensor([local_pos[c.id()] for c in fusable],
-                                   device=self.device)
-                U_local.index_copy_(0, idx, Uf)
-            if custom:
-                updates = self._loop.run_round(self.global_model, custom,
-                                               self.dataset, local_steps, lr)
-                for c in custom:
-                    U_local[local_pos[c.id()]].copy_(
-                        updates[c.id()].to(self.device))
+            if len(fusable) == len(shard):
+                self._fused.run_round(theta, shard, self.dataset,
+                                      local_steps, lr, out=U_local)
+            else:
+                if fusable:
+                    Uf = self._fused.run_round(theta, fusable, self.dataset,
+                                               local_steps, lr)
+                    idx = torch.tensor([local_pos[c.id()] for c in fusable],
+                                       device=self.device)
+                    U_local.index_copy_(0, idx, Uf)
+                if custom:
+                    updates = self._loop.run_round(self.global_model, custom,
+                                                   self.dataset, local_steps, lr)
+                    for c in custom:
+                        U_local[local_pos[c.id()]].copy_(
+                            updates[c.id()].to(self.device))
 
         with trace_range("blades/gather"):
-            U = rt.all_gather_rows(U_local, total_rows=len(all_clients))
+            buf = rt.all_gather_rows(buf_local, total_rows=len(all_clients))
+            U = buf[:, :d]
             torch.nan_to_num_(U)  # K18 sanitize (reference: client.py:198)
 
         # hand every client its row view (zero-copy)

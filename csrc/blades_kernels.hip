@@ -154,7 +154,11 @@ __global__ void trimmed_select_kernel(const float* __restrict__ U,
   const float inv = 1.f / (float)(K - 2 * b);
 
   for (long long j = tid; j < d; j += nthreads) {
-    float sum = 0.f;
+    // fp64 running/trim sums: the result is (sum − trim) where the two can
+    // cancel to ~(K−2b)/K of their magnitude (median: 2 of K survive) —
+    // fp32 here loses ~1e-6 absolute; fp64 makes the subtraction exact at
+    // fp32 output precision.  The kernel stays memory-bound.
+    double sum = 0.0;
     int nlo = 0, nhi = 0;
     float lo_max = -INFINITY, hi_min = INFINITY;
     int lo_max_i = 0, hi_min_i = 0;
@@ -191,9 +195,10 @@ __global__ void trimmed_select_kernel(const float* __restrict__ U,
         }
       }
     }
-    float trim = 0.f;
-    for (int i = 0; i < b; ++i) trim += lo[i * BS] + hi[i * BS];
-    out[j] = (sum - trim) * inv;
+    double trim = 0.0;
+    for (int i = 0; i < b; ++i)
+      trim += (double)lo[i * BS] + (double)hi[i * BS];
+    out[j] = (float)((sum - trim) * (double)inv);
   }
 }
 
