@@ -47,21 +47,24 @@ class SyntheticFLDataset:
         return self._clients
 
     def _gen_pool(self, u_id: int, n: int, tag: int):
-        g = torch.Generator(device=self.device)
+        # ALWAYS generate on CPU: CUDA and CPU philox streams differ, and a
+        # client's data must be identical regardless of the device (or rank)
+        # that hosts it.  One-time cost outside any timed region; pools then
+        # live on self.device.
+        g = torch.Generator()
         g.manual_seed(client_philox_seed(self.seed, int(u_id), 0, tag=tag))
-        X = torch.randn((n, *self.shape), generator=g, device=self.device)
+        X = torch.randn((n, *self.shape), generator=g)
         if self.learnable:
             if self._teacher is None:
-                gt = torch.Generator(device=self.device)
+                gt = torch.Generator()
                 gt.manual_seed(client_philox_seed(self.seed, 0, 0, tag=99))
                 dim = int(torch.tensor(self.shape).prod())
                 self._teacher = torch.randn((dim, self.num_classes),
-                                            generator=gt, device=self.device)
+                                            generator=gt)
             y = (X.flatten(1) @ self._teacher).argmax(dim=1)
         else:
-            y = torch.randint(0, self.num_classes, (n,), generator=g,
-                              device=self.device)
-        return X, y
+            y = torch.randint(0, self.num_classes, (n,), generator=g)
+        return X.to(self.device), y.to(self.device)
 
     def _train_pool(self, u_id: int):
         if u_id not in self._pool:
