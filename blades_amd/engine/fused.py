@@ -103,6 +103,16 @@ class FusedEngine:
 
     @torch.no_grad()
     def _client_vectors(self, clients: Sequence[BladesClient]):
+        key = tuple(id(c) for c in clients)
+        cached = getattr(self, "_cv_cache", None)
+        if cached is not None and cached[0] == key:
+            return cached[1]
+        result = self._client_vectors_uncached(clients)
+        self._cv_cache = (key, result)
+        return result
+
+    @torch.no_grad()
+    def _client_vectors_uncached(self, clients: Sequence[BladesClient]):
         C = len(clients)
         clamp_hi = torch.full((C,), 1e6, device=self.device)
         grad_sign = torch.ones((C,), device=self.device)
@@ -128,27 +138,35 @@ class FusedEngine:
         return X, Y
 
     def run_round(self, theta: Tensor, clients: List[BladesClient], dataset,
-                  local_steps: int, lr: float,
-                  out: Optional[Tensor] = None) -> Tensor:
+                  local_steps: int, lr,
+                  out: Optional[Tensor] = None, data=None) -> Tensor:
         """Returns the update slab U = θ_after − θ [C, d] on ``self.device``.
 
         ``theta`` is the flat global parameter vector (device-resident).
         ``out``: optional [C, d] destination (may be a padded-row view) the
         updates are written into directly — saves a slab copy per round.
+        ``lr`` may be a float or a 0-dim device tensor (hipGraph capture
+        keeps learning rates in device scalars so schedulers need no
+        re-capture).  ``data``: pre-staged [(X [C,B,...], Y [C,B])] per
+        local step — bypasses the dataset (the graph path owns static
+        buffers).
         """
         C = len(clients)
         clamp_hi, grad_sign, target_tfms = self._client_vectors(clients)
 
-        # fetch all batches up front (synthetic/device datasets return views)
-        stacked = getattr(dataset, "get_stacked_train_data", None)
-        if stacked is not None:
-            steps_data = stacked([c.id() for c in clients], local_steps,
-                                 device=self.device)
+        if data is not None:
+            steps_data = data
         else:
-            per_client = {c.id(): dataset.get_train_data(c.id(), local_steps)
-                          for c in clients}
-            steps_data = [self._stack_step(dataset, clients, s, per_client)
-                          for s in range(local_steps)]
+            # fetch all batches up front (device datasets return views)
+            stacked = getattr(dataset, "get_stacked_train_data", None)
+            if stacked is not None:
+                steps_data = stacked([c.id() for c in clients], local_steps,
+                                     device=self.device)
+            else:
+                per_client = {c.id(): dataset.get_train_data(c.id(), local_steps)
+                              for c in clients}
+                steps_data = [self._stack_step(dataset, clients, s, per_client)
+                              for s in range(local_steps)]
 
         # apply fused target transforms once per stacked step
         byz_tt_rows = [i for i, t in enumerate(target_tfms) if t is not None]
@@ -187,8 +205,13 @@ class FusedEngine:
             params = tuple(slab_views[n] for n in self.spec.names)
             grads = grad_fn(params, X, Y, clamp_hi, grad_sign)
             with torch.no_grad():
-                torch._foreach_add_([slab_views[n] for n in self.spec.names],
-                                    list(grads), alpha=-lr)
+                views = [slab_views[n] for n in self.spec.names]
+                if isinstance(lr, torch.Tensor):
+                    gl = list(grads)
+                    torch._foreach_mul_(gl, -lr)
+                    torch._foreach_add_(views, gl)
+                else:
+                    torch._foreach_add_(views, list(grads), alpha=-lr)
         slab.sub_(theta.unsqueeze(0))
         if out is not None:
             out.copy_(slab)

@@ -66,8 +66,8 @@ class BladesServer:
         self.optimizer.step()
 
     # ------- flat-vector access used by the distributed runtime/checkpoint
-    def flat_parameters(self, device=None) -> torch.Tensor:
-        return self._spec.flatten(self.model, device=device)
+    def flat_parameters(self, device=None, out=None) -> torch.Tensor:
+        return self._spec.flatten(self.model, device=device, out=out)
 
     def load_flat_parameters(self, vec: torch.Tensor) -> None:
         self._spec.load(self.model, vec)

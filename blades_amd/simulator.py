@@ -87,6 +87,7 @@ class Simulator:
         engine: str = "auto",
         client_chunk: Optional[int] = None,
         device: Optional[str] = None,
+        hip_graphs: bool = True,
         **kwargs,
     ):
         self.use_actor = mode == "actor"
@@ -149,6 +150,15 @@ class Simulator:
         self._fused: Optional[FusedEngine] = None
         self._loop: Optional[LoopEngine] = None
         self.global_model: Optional[torch.nn.Module] = None
+        # hipGraph round capture (engine/graphs.py); auto-disabled when the
+        # population/aggregator/optimizer is not capturable
+        import os as _os
+
+        self._use_graphs = hip_graphs and _os.environ.get(
+            "BLADES_AMD_NO_GRAPHS", "0") != "1"
+        self._graph_round = None
+        self._theta: Optional[torch.Tensor] = None
+        self._model_stale = False  # True while θ (graph mode) is ahead of model
 
     # ------------------------------------------------------------ builders
     def _init_aggregator(self, aggregator, aggregator_kws) -> None:
@@ -249,9 +259,45 @@ class Simulator:
             client.device = self.device
             client.set_model(self.global_model, torch.optim.SGD, lr)
 
+    def sync_model(self) -> None:
+        """Write θ (graph-mode source of truth) back into the module params."""
+        if self._model_stale:
+            self.server.load_flat_parameters(self._theta)
+            self._model_stale = False
+
+    def _maybe_graph_round(self, global_round: int, local_steps: int,
+                           lr: float) -> bool:
+        if not self._use_graphs or self.device.type != "cuda":
+            return False
+        from blades_amd.engine.graphs import CapturedRound
+
+        if self._graph_round is None:
+            reason = CapturedRound.supported(self, self.get_clients(),
+                                             local_steps)
+            if reason is not None:
+                self.debug_logger.info(f"hipGraph round capture off: {reason}")
+                self._use_graphs = False
+                return False
+            self.server.flat_parameters(device=self.device, out=self._theta)
+            self._graph_round = CapturedRound(self, self.get_clients(),
+                                              local_steps)
+        if self._graph_round.local_steps != local_steps:
+            self.debug_logger.info("local_steps changed; dropping hipGraph")
+            self._graph_round = None
+            self._use_graphs = False
+            return False
+        server_lr = self.server_opt.param_groups[0]["lr"]
+        with trace_range("blades/graph_round"):
+            self._graph_round.run(lr, server_lr)
+        self._model_stale = True
+        return True
+
     def train_round(self, global_round: int, local_steps: int,
                     clients: List[BladesClient], lr: float) -> None:
         """One global round (reference: train_actor, simulator.py:203-247)."""
+        if self._maybe_graph_round(global_round, local_steps, lr):
+            return
+        self.sync_model()
         rt = self.runtime
         all_clients = self.get_clients()
         rows = {c.id(): i for i, c in enumerate(all_clients)}
@@ -262,7 +308,8 @@ class Simulator:
         d = self._spec.d
         d_pad = -(-d // 4) * 4
         with trace_range("blades/local_train"):
-            theta = self.server.flat_parameters(device=self.device)
+            theta = self.server.flat_parameters(device=self.device,
+                                                out=self._theta)
             fusable, custom = split_fusable(shard)
             if self._engine_choice == "loop":
                 fusable, custom = [], shard
@@ -335,11 +382,16 @@ class Simulator:
         shard = rt.my_shard(all_clients)
         with trace_range("blades/eval"):
             if self._fused is not None and self._engine_choice != "loop":
-                theta = self.server.flat_parameters(device=self.device)
+                if self._model_stale:
+                    theta = self._theta  # graph mode: θ is current
+                else:
+                    theta = self.server.flat_parameters(device=self.device,
+                                                        out=self._theta)
                 local_metrics = self._fused.evaluate(
                     theta, shard, self.dataset, global_round, batch_size,
                     self.metrics)
             else:
+                self.sync_model()
                 for c in shard:
                     self._ensure_client_model(c, 0.0)
                 local_metrics = self._loop.evaluate(
@@ -404,6 +456,9 @@ class Simulator:
         model = model.to(self.device)
         self.global_model = model
         self._spec = ParamSpec.from_module(model)
+        self._theta = torch.empty(self._spec.d, device=self.device)
+        self._graph_round = None
+        self._model_stale = False
 
         if server_optimizer == "SGD":
             self.server_opt = torch.optim.SGD(model.parameters(), lr=server_lr)
@@ -453,6 +508,7 @@ class Simulator:
             self.debug_logger.info(
                 f"E={r}; Client learning rate = {cur_lr}; "
                 f"Time cost = {time.time() - global_start}")
+        self.sync_model()
         return ret
 
     def __str__(self) -> str:
