@@ -102,6 +102,7 @@ class CapturedRound:
         # static attack plan
         self.honest_mask = torch.tensor(
             [not c.is_byzantine() for c in clients], device=self.device)
+        self.n_honest = int(self.honest_mask.sum().item())
         alie_groups = {}
         ipm_groups = {}
         for i, c in enumerate(clients):
@@ -132,11 +133,13 @@ class CapturedRound:
         torch.nan_to_num_(self.U)
         for z, rows in self.alie_groups:
             mu, std = ops.masked_col_mean_std(self.U, self.honest_mask,
-                                              unbiased=True)
+                                              unbiased=True,
+                                              count=self.n_honest)
             self.U.index_copy_(0, rows,
                                (mu - std * z).unsqueeze(0).expand(len(rows), -1))
         for eps, rows in self.ipm_groups:
-            hm = ops.masked_col_mean(self.U, self.honest_mask)
+            hm = ops.masked_col_mean(self.U, self.honest_mask,
+                                     count=self.n_honest)
             self.U.index_copy_(0, rows,
                                (-eps * hm).unsqueeze(0).expand(len(rows), -1))
         delta = sim.aggregator(self.U)

@@ -113,19 +113,24 @@ def weighted_col_sum(U: Tensor, w: Tensor) -> Tensor:
     return _ref.weighted_col_sum(U, w)
 
 
-def masked_col_mean(U: Tensor, mask: Tensor) -> Tensor:
+def masked_col_mean(U: Tensor, mask: Tensor, count: int = -1) -> Tensor:
+    """``count`` = number of True rows; pass it explicitly from inside a
+    hipGraph capture (computing it would sync the stream)."""
     ext = _route(U)
     if ext is not None:
-        return ext.masked_col_mean(_prep(U), mask.to(torch.bool).contiguous())
+        return ext.masked_col_mean(_prep(U), mask.to(torch.bool).contiguous(),
+                                   float(count))
     return _ref.masked_col_mean(U, mask)
 
 
-def masked_col_mean_std(U: Tensor, mask: Tensor,
-                        unbiased: bool = True) -> Tuple[Tensor, Tensor]:
+def masked_col_mean_std(U: Tensor, mask: Tensor, unbiased: bool = True,
+                        count: int = -1) -> Tuple[Tensor, Tensor]:
+    """``count`` as in :func:`masked_col_mean`."""
     ext = _route(U)
     if ext is not None:
         return ext.masked_col_mean_std(_prep(U),
-                                       mask.to(torch.bool).contiguous(), unbiased)
+                                       mask.to(torch.bool).contiguous(),
+                                       unbiased, float(count))
     return _ref.masked_col_mean_std(U, mask, unbiased)
 
 

@@ -410,11 +410,14 @@ torch::Tensor weighted_col_sum(torch::Tensor U, torch::Tensor w) {
   return out;
 }
 
-torch::Tensor masked_col_mean(torch::Tensor U, torch::Tensor mask) {
+torch::Tensor masked_col_mean(torch::Tensor U, torch::Tensor mask,
+                              double count) {
   auto v = view_of(U);
   TORCH_CHECK(mask.scalar_type() == at::kBool && mask.numel() == v.K);
   auto mc = mask.contiguous();
-  const float cnt = (float)mask.sum().item<long>();
+  // count passed by the caller: .item() here would sync the stream, which
+  // is illegal under hipGraph capture (the mask is static in that regime)
+  const float cnt = (float)(count >= 0 ? count : mask.sum().item<long>());
   auto out = torch::empty({v.d}, U.options());
   const int BS = 256;
   auto stream = c10::hip::getCurrentHIPStream().stream();
@@ -431,11 +434,12 @@ torch::Tensor masked_col_mean(torch::Tensor U, torch::Tensor mask) {
 }
 
 std::tuple<torch::Tensor, torch::Tensor>
-masked_col_mean_std(torch::Tensor U, torch::Tensor mask, bool unbiased) {
+masked_col_mean_std(torch::Tensor U, torch::Tensor mask, bool unbiased,
+                    double count) {
   auto v = view_of(U);
   TORCH_CHECK(mask.scalar_type() == at::kBool && mask.numel() == v.K);
   auto mc = mask.contiguous();
-  const float cnt = (float)mask.sum().item<long>();
+  const float cnt = (float)(count >= 0 ? count : mask.sum().item<long>());
   TORCH_CHECK(cnt >= 2, "need >=2 masked rows for std");
   auto mu = torch::empty({v.d}, U.options());
   auto sd = torch::empty({v.d}, U.options());
@@ -548,9 +552,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "blades_amd CDNA4 HIP kernels (gfx950)";
   m.def("col_mean", &col_mean, "column mean (K1)");
   m.def("weighted_col_sum", &weighted_col_sum, "weighted column sum");
-  m.def("masked_col_mean", &masked_col_mean, "masked column mean (K11)");
+  m.def("masked_col_mean", &masked_col_mean, "masked column mean (K11)",
+        py::arg("U"), py::arg("mask"), py::arg("count") = -1.0);
   m.def("masked_col_mean_std", &masked_col_mean_std,
-        "masked column mean+std (K10)");
+        "masked column mean+std (K10)", py::arg("U"), py::arg("mask"),
+        py::arg("unbiased") = true, py::arg("count") = -1.0);
   m.def("trimmed_mean", &trimmed_mean, "coordinate-wise trimmed mean (K3)");
   m.def("col_median", &col_median, "coordinate-wise median (K2)");
   m.def("row_sq_norms", &row_sq_norms, "per-row squared norms");
