@@ -67,6 +67,7 @@ class FusedEngine:
     def __init__(self, model: nn.Module, spec: ParamSpec, device,
                  client_chunk: Optional[int] = None):
         import copy
+        import os
 
         self.device = torch.device(device)
         self.spec = spec
@@ -76,6 +77,16 @@ class FusedEngine:
         self.buffers = {k: v.detach().clone() for k, v in self.base.named_buffers()}
         self.client_chunk = client_chunk
         self._grad_fn = None
+        # population path: direct MFMA popconv kernels for supported models
+        # (GPU; CPU uses it only when forced, for parity tests)
+        from blades_amd.engine.popmodel import population_forward_for
+
+        self._pop_fn = None
+        if os.environ.get("BLADES_AMD_NO_POPCONV", "0") != "1":
+            fn = population_forward_for(model)
+            if fn is not None and (self.device.type == "cuda"
+                                   or os.environ.get("BLADES_AMD_FORCE_POP") == "1"):
+                self._pop_fn = fn
 
     # ------------------------------------------------------------ internals
     def _loss(self, params_tuple: Tuple[Tensor, ...], x: Tensor, y: Tensor,
@@ -179,6 +190,10 @@ class FusedEngine:
                 Y[i] = target_tfms[i](Y[i])
             return Y
 
+        if self._pop_fn is not None:
+            return self._pop_round(theta, clients, steps_data, local_steps,
+                                   lr, out, clamp_hi, grad_sign, fix_targets)
+
         fedsgd = local_steps == 1
         if fedsgd:
             grad_fn = self._build_grad_fn(shared_params=True)
@@ -212,6 +227,72 @@ class FusedEngine:
                     torch._foreach_add_(views, gl)
                 else:
                     torch._foreach_add_(views, list(grads), alpha=-lr)
+        slab.sub_(theta.unsqueeze(0))
+        if out is not None:
+            out.copy_(slab)
+            return out
+        return slab
+
+    # --------------------------------------------------- population path
+    def _pop_round(self, theta: Tensor, clients, steps_data, local_steps: int,
+                   lr, out: Optional[Tensor], clamp_hi: Tensor,
+                   grad_sign: Tensor, fix_targets) -> Tensor:
+        """Direct-MFMA population training (engine/popmodel.py): explicit
+        autograd over batched [C, *shape] parameters instead of vmap.
+
+        FedSGD (shared θ): parameters enter the graph as stride-0 expands of
+        the flat θ views — popconv reads them broadcast, autograd.grad
+        w.r.t. the EXPANDED tensors returns dense per-client gradients.
+        FedAvg: parameters are detached slab views updated in place.
+        """
+        import torch.nn.functional as F
+
+        C = len(clients)
+        fedsgd = local_steps == 1
+        names = self.spec.names
+
+        def per_client_loss_sum(params, X, Y):
+            logits = self._pop_fn(params, X)  # [C, B, nc]
+            nc = logits.shape[-1]
+            losses = F.cross_entropy(logits.reshape(-1, nc), Y.reshape(-1),
+                                     reduction="none").view(C, -1).mean(1)
+            losses = torch.minimum(torch.clamp(losses, min=0.0), clamp_hi)
+            return (losses * grad_sign).sum()
+
+        if fedsgd:
+            base = dict(self.spec.named_slices(theta))
+            params = {
+                n: base[n].unsqueeze(0).expand(C, *base[n].shape)
+                .detach().requires_grad_() for n in names
+            }
+            X, Y = steps_data[0]
+            loss = per_client_loss_sum(params, X, fix_targets(Y))
+            grads = torch.autograd.grad(loss, [params[n] for n in names])
+            U = out if out is not None else torch.empty(
+                (C, self.spec.d), device=self.device)
+            views = self.spec.batched_views(U)
+            with torch.no_grad():
+                for n, g in zip(names, grads):
+                    views[n].copy_(g)
+            U.mul_(-lr)
+            return U
+
+        slab = theta.unsqueeze(0).repeat(C, 1).contiguous()
+        slab_views = self.spec.batched_views(slab)
+        for s in range(local_steps):
+            params = {n: slab_views[n].detach().requires_grad_()
+                      for n in names}
+            X, Y = steps_data[s]
+            loss = per_client_loss_sum(params, X, fix_targets(Y))
+            grads = torch.autograd.grad(loss, [params[n] for n in names])
+            with torch.no_grad():
+                gl = list(grads)
+                if isinstance(lr, torch.Tensor):
+                    torch._foreach_mul_(gl, -lr)
+                    torch._foreach_add_([slab_views[n] for n in names], gl)
+                else:
+                    torch._foreach_add_([slab_views[n] for n in names], gl,
+                                        alpha=-lr)
         slab.sub_(theta.unsqueeze(0))
         if out is not None:
             out.copy_(slab)

@@ -1,0 +1,177 @@
+"""Population-conv model: CPU parity vs the standard per-client model, and
+(on GPU) MFMA kernel parity vs the CPU reference."""
+import os
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from blades_amd.engine.flat import ParamSpec
+from blades_amd.engine.popmodel import (_PopConv3x3, interior, pop_batchnorm,
+                                        pop_conv3x3, pop_resnet18,
+                                        to_pop_layout)
+from blades_amd.models import resnet18
+
+
+def expand_params(model, C):
+    return {n: p.detach().unsqueeze(0).expand(C, *p.shape)
+            for n, p in model.named_parameters()}
+
+
+def test_popconv_cpu_forward_matches_conv2d():
+    torch.manual_seed(0)
+    C, ci, co, B, H = 3, 4, 8, 2, 6
+    x = torch.randn(C, B, ci, H, H)
+    w = torch.randn(C, co, ci, 3, 3)
+    xp = to_pop_layout(x)
+    y = _PopConv3x3.apply(xp, w)
+    for c in range(C):
+        ref = F.conv2d(x[c], w[c], padding=1)
+        got = interior(y[c]).permute(1, 0, 2, 3)
+        assert torch.allclose(got, ref, atol=1e-5), c
+
+
+def test_popconv_cpu_backward_matches_autograd():
+    torch.manual_seed(1)
+    C, ci, co, B, H = 2, 4, 4, 2, 5
+    x = torch.randn(C, B, ci, H, H)
+    w = torch.randn(C, co, ci, 3, 3, requires_grad=True)
+    xp = to_pop_layout(x).requires_grad_()
+    y = _PopConv3x3.apply(xp, w)
+    loss = (y ** 2).sum()
+    gx, gw = torch.autograd.grad(loss, [xp, w])
+
+    # reference: per-client autograd through F.conv2d
+    x2 = x.clone().requires_grad_()
+    w2 = w.detach().clone().requires_grad_()
+    loss2 = 0
+    for c in range(C):
+        yc = F.conv2d(x2[c], w2[c], padding=1)
+        loss2 = loss2 + (yc ** 2).sum()
+    gx2, gw2 = torch.autograd.grad(loss2, [x2, w2])
+    assert torch.allclose(gw, gw2, atol=1e-4)
+    got = interior(gx).permute(0, 2, 1, 3, 4)
+    assert torch.allclose(got, gx2, atol=1e-4)
+
+
+def test_pop_resnet18_forward_matches_module():
+    torch.manual_seed(2)
+    C, B = 3, 4
+    m = resnet18(norm="batch-local").eval()
+    x = torch.randn(C, B, 3, 32, 32)
+    params = expand_params(m, C)
+    out = pop_resnet18(params, x)
+    assert out.shape == (C, B, 10)
+    for c in range(C):
+        ref = m(x[c])
+        assert torch.allclose(out[c], ref, atol=1e-4), (
+            c, (out[c] - ref).abs().max())
+
+
+def test_pop_engine_matches_loop_engine():
+    """Full engine on the population path (forced on CPU) vs loop engine."""
+    os.environ["BLADES_AMD_FORCE_POP"] = "1"
+    try:
+        from blades_amd import Simulator
+        from blades_amd.datasets import SyntheticFLDataset
+
+        def run(engine):
+            torch.manual_seed(3)
+            ds = SyntheticFLDataset(num_clients=4, samples_per_client=8,
+                                    batch_size=4, shape=(3, 32, 32),
+                                    num_classes=10, seed=0)
+            sim = Simulator(ds, num_byzantine=1, attack="signflipping",
+                            aggregator="mean",
+                            log_path=f"/tmp/bl_pop_{engine}", seed=5,
+                            engine=engine)
+            sim.run(resnet18(norm="batch-local"), global_rounds=2,
+                    local_steps=2, client_lr=0.05, server_lr=1.0,
+                    validate_interval=0)
+            return sim.server.flat_parameters()
+
+        a = run("auto")   # population path (forced)
+        b = run("loop")   # reference per-client semantics
+        # popconv reduces in a different fp32 order than F.conv2d; the
+        # per-round drift is ~2e-5 max and compounds over rounds/steps
+        assert torch.allclose(a, b, atol=5e-3)
+        assert (a - b).abs().mean() < 1e-5
+    finally:
+        os.environ.pop("BLADES_AMD_FORCE_POP", None)
+
+
+@pytest.mark.gpu
+def test_popconv_kernels_match_cpu():
+    torch.manual_seed(4)
+    for C, ci, co, B, H in [(3, 4, 32, 2, 8), (2, 32, 64, 4, 16),
+                            (5, 3, 64, 2, 32), (2, 64, 64, 8, 32)]:
+        x = torch.randn(C, B, ci, H, H)
+        w = torch.randn(C, co, ci, 3, 3)
+        xp = to_pop_layout(x)
+        y_cpu = _PopConv3x3.apply(xp, w)
+        y_gpu = _PopConv3x3.apply(xp.cuda().contiguous(), w.cuda())
+        assert torch.allclose(y_gpu.cpu(), y_cpu, atol=1e-3,
+                              rtol=1e-4), (C, ci, co)
+
+        # backward parity
+        xp_g = xp.cuda().contiguous().requires_grad_()
+        w_g = w.cuda().requires_grad_()
+        gx_g, gw_g = torch.autograd.grad(
+            (_PopConv3x3.apply(xp_g, w_g) ** 2).sum(), [xp_g, w_g])
+        xp_c = xp.clone().requires_grad_()
+        w_c = w.clone().requires_grad_()
+        gx_c, gw_c = torch.autograd.grad(
+            (_PopConv3x3.apply(xp_c, w_c) ** 2).sum(), [xp_c, w_c])
+        assert torch.allclose(gw_g.cpu(), gw_c, atol=5e-2, rtol=1e-3)
+        assert torch.allclose(interior(gx_g).cpu(), interior(gx_c),
+                              atol=1e-2, rtol=1e-3)
+
+
+@pytest.mark.gpu
+def test_popconv_shared_weight_broadcast():
+    torch.manual_seed(5)
+    C, ci, co, B, H = 4, 32, 32, 4, 16
+    x = torch.randn(C, B, ci, H, H, device="cuda")
+    w1 = torch.randn(co, ci, 3, 3, device="cuda")
+    w = w1.unsqueeze(0).expand(C, co, ci, 3, 3)
+    xp = to_pop_layout(x).contiguous()
+    y = _PopConv3x3.apply(xp, w)
+    y2 = _PopConv3x3.apply(xp, w.contiguous())
+    assert torch.allclose(y, y2, atol=1e-4)
+
+
+@pytest.mark.gpu
+def test_pop_engine_gpu_matches_vmap():
+    import os as _os
+
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    def run(no_pop):
+        env = dict(BLADES_AMD_NO_POPCONV="1") if no_pop else {}
+        old = {k: _os.environ.get(k) for k in env}
+        _os.environ.update(env)
+        try:
+            torch.manual_seed(6)
+            ds = SyntheticFLDataset(num_clients=8, samples_per_client=16,
+                                    batch_size=8, shape=(3, 32, 32),
+                                    num_classes=10, seed=0, device="cuda:0")
+            sim = Simulator(ds, num_byzantine=2, attack="alie",
+                            attack_kws={"num_clients": 8, "num_byzantine": 2},
+                            aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                            use_cuda=True, log_path=f"/tmp/bl_popg_{no_pop}",
+                            seed=7, hip_graphs=False)
+            sim.run(resnet18(norm="batch-local"), global_rounds=2,
+                    local_steps=1, client_lr=0.1, server_lr=1.0,
+                    validate_interval=0)
+            return sim.server.flat_parameters().cpu()
+        finally:
+            for k, v in old.items():
+                if v is None:
+                    _os.environ.pop(k, None)
+                else:
+                    _os.environ[k] = v
+
+    a = run(False)
+    b = run(True)
+    assert torch.allclose(a, b, atol=1e-3)
+    assert (a - b).abs().mean() < 5e-6
