@@ -85,16 +85,18 @@ def test_pop_engine_matches_loop_engine():
                             log_path=f"/tmp/bl_pop_{engine}", seed=5,
                             engine=engine)
             sim.run(resnet18(norm="batch-local"), global_rounds=2,
-                    local_steps=2, client_lr=0.05, server_lr=1.0,
+                    local_steps=2, client_lr=0.005, server_lr=1.0,
                     validate_interval=0)
             return sim.server.flat_parameters()
 
         a = run("auto")   # population path (forced)
         b = run("loop")   # reference per-client semantics
-        # popconv reduces in a different fp32 order than F.conv2d; the
-        # per-round drift is ~2e-5 max and compounds over rounds/steps
-        assert torch.allclose(a, b, atol=5e-3)
-        assert (a - b).abs().mean() < 1e-5
+        # popconv reduces in a different fp32 order than F.conv2d (~2e-5
+        # per step); at large lr the BN Jacobians of a random-init ResNet
+        # amplify that chaotically (verified: lr 0.05 -> 1e-2, lr 1e-3 ->
+        # 1e-5), so this parity test runs at a small, stable lr
+        assert torch.allclose(a, b, atol=5e-4)
+        assert (a - b).abs().mean() < 2e-6
     finally:
         os.environ.pop("BLADES_AMD_FORCE_POP", None)
 
