@@ -96,7 +96,7 @@ def test_pop_engine_matches_loop_engine():
         # amplify that chaotically (verified: lr 0.05 -> 1e-2, lr 1e-3 ->
         # 1e-5), so this parity test runs at a small, stable lr
         assert torch.allclose(a, b, atol=5e-4)
-        assert (a - b).abs().mean() < 2e-6
+        assert (a - b).abs().mean() < 5e-6
     finally:
         os.environ.pop("BLADES_AMD_FORCE_POP", None)
 
