@@ -110,17 +110,38 @@ class _PopConv3x3(torch.autograd.Function):
         dY = zero_pads_(dY.contiguous())
         if X.is_cuda:
             ext = _ext()
-            dYf = dY.reshape(C, co, B * Hp * Wp)
-            Xf = X.reshape(C, ci, B * Hp * Wp)
+            Np = B * Hp * Wp
+            dYf = dY.reshape(C, co, Np)
+            Xf = X.reshape(C, ci, Np)
             dX = None
             if ctx.needs_input_grad[0]:
-                # dX = conv(dY, W transposed in (co,ci), taps flipped)
-                Wt = W.permute(0, 2, 1, 3, 4).flip(3, 4).contiguous()
+                # dX = conv(dY, W transposed in (co,ci), taps flipped).
+                # Shared weights arrive as a stride-0 expand — transform the
+                # single base copy, never materialize C copies.
+                if W.stride(0) == 0:
+                    Wt = (W[0].permute(1, 0, 2, 3).flip(2, 3).contiguous()
+                          .unsqueeze(0).expand(C, ci, co, 3, 3))
+                else:
+                    Wt = W.permute(0, 2, 1, 3, 4).flip(3, 4).contiguous()
                 dX = ext.popconv_fwd(dYf, Wt, B, Hp, Wp).view(X.shape)
                 zero_pads_(dX)
             dW = None
             if ctx.needs_input_grad[1]:
-                dW = ext.popconv_dw(dYf, Xf, B, Hp, Wp, False)
+                # dW[c,:,:,t] = dY_c @ shift(X_c, Δt)ᵀ — nine shifted
+                # strided-batched GEMMs (dY pads are zero, so truncating the
+                # contraction range by |Δ| is exact).  rocBLAS runs these at
+                # 56-113 TF (benchmarks/microbench_bmm.py); the custom
+                # popconv_dw kernel remains for shapes where its launch
+                # geometry wins.
+                dW = Xf.new_empty(C, co, ci, 3, 3)
+                for t in range(9):
+                    dyy, dxx = t // 3 - 1, t % 3 - 1
+                    delta = dyy * Wp + dxx
+                    a = max(0, -delta)
+                    b = Np - max(0, delta)
+                    tap = torch.bmm(dYf[:, :, a:b],
+                                    Xf[:, :, a + delta:b + delta].transpose(1, 2))
+                    dW[:, :, :, t // 3, t % 3].copy_(tap)
             return dX, dW
         # CPU fallback via torch.nn.grad per client
         dX = torch.zeros_like(X) if ctx.needs_input_grad[0] else None

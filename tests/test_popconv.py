@@ -162,8 +162,10 @@ def test_pop_engine_gpu_matches_vmap():
                             aggregator="trimmedmean", aggregator_kws={"nb": 2},
                             use_cuda=True, log_path=f"/tmp/bl_popg_{no_pop}",
                             seed=7, hip_graphs=False)
+            # small lr: BN-Jacobian chaos amplifies fp32 reduction-order
+            # noise at high lr (see test_pop_engine_matches_loop_engine)
             sim.run(resnet18(norm="batch-local"), global_rounds=2,
-                    local_steps=1, client_lr=0.1, server_lr=1.0,
+                    local_steps=1, client_lr=0.01, server_lr=1.0,
                     validate_interval=0)
             return sim.server.flat_parameters().cpu()
         finally:
