@@ -44,7 +44,7 @@ namespace {
 
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
-constexpr int NT = 128;        // n-tile per block (4 waves × 32)
+constexpr int NT = 256;        // n-tile per block (4 waves × 2×32)
 constexpr int CK = 32;         // ci chunk
 constexpr int MAX_WP = 40;     // max padded width (CIFAR: 34)
 constexpr int XROW = NT + 2 * MAX_WP + 2 + 2;  // LDS X row length (+2 bank skew)
@@ -59,8 +59,11 @@ void popconv_fwd_kernel(const float* __restrict__ X,
                         float* __restrict__ Y,
                         int C, int co, int ci, long long Np, int Wp,
                         long long strideWc /* 0 when shared */) {
+  // LDS: X chunk with halo (CK x XROW ≈ 42 KiB) + all nine taps' weights
+  // for the chunk (9 x 32 x 33 ≈ 38 KiB) -> 2 barriers per ci chunk
+  // instead of 2 per (chunk, tap).
   __shared__ float Xs[CK][XROW];
-  __shared__ float Ws[32][33];
+  __shared__ float Ws[9][32][33];
 
   const int c = blockIdx.z;
   const int co0 = blockIdx.y * 32;
@@ -76,12 +79,12 @@ void popconv_fwd_kernel(const float* __restrict__ X,
   const int arow = lane & 31;       // A row (co) / B col (n)
   const int ksel = lane >> 5;       // which of the 2 contraction cols
 
-  f32x16 acc = {};
+  f32x16 acc0 = {};                 // wave's n sub-tiles: wid*64 + {0,32}
+  f32x16 acc1 = {};
 
   const int n_chunks = (ci + CK - 1) / CK;
   for (int cc = 0; cc < n_chunks; ++cc) {
     const int ci0 = cc * CK;
-    // ---- stage X chunk with halo: rows = CK ci, cols = NT + 2*halo
     __syncthreads();
     {
       const int cols = NT + 2 * halo;
@@ -92,33 +95,37 @@ void popconv_fwd_kernel(const float* __restrict__ X,
         Xs[r][q] = (cii < ci && n >= 0 && n < Np)
                        ? Xc[(long long)cii * Np + n] : 0.f;
       }
+      for (int e = threadIdx.x; e < 9 * 32 * CK; e += blockDim.x) {
+        const int t = e / (32 * CK);
+        const int r = (e / CK) % 32, k = e % CK;
+        const int coo = co0 + r, cii = ci0 + k;
+        Ws[t][r][k] = (coo < co && cii < ci)
+                          ? Wc[(((long long)coo * ci + cii) * 9) + t] : 0.f;
+      }
     }
+    __syncthreads();
+    #pragma unroll
     for (int t = 0; t < 9; ++t) {
       const int dy = t / 3 - 1, dx = t % 3 - 1;
       const int delta = dy * Wp + dx;
-      // ---- stage W[co0:co0+32, ci0:ci0+CK, t]
-      __syncthreads();
-      for (int e = threadIdx.x; e < 32 * CK; e += blockDim.x) {
-        const int r = e / CK, k = e % CK;
-        const int coo = co0 + r, cii = ci0 + k;
-        Ws[r][k] = (coo < co && cii < ci)
-                       ? Wc[(((long long)coo * ci + cii) * 9) + t] : 0.f;
-      }
-      __syncthreads();
-      // ---- 16 MFMA steps over the CK=32 contraction
-      const int nn = halo + delta + wid * 32 + arow;
+      const int nn = halo + delta + wid * 64 + arow;
       #pragma unroll
       for (int k2 = 0; k2 < CK / 2; ++k2) {
-        const float a = Ws[arow][2 * k2 + ksel];
-        const float b = Xs[2 * k2 + ksel][nn];
-        acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+        const float a = Ws[t][arow][2 * k2 + ksel];
+        const float b0 = Xs[2 * k2 + ksel][nn];
+        const float b1 = Xs[2 * k2 + ksel][nn + 32];
+        acc0 = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b0, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b1, acc1, 0, 0, 0);
       }
     }
   }
 
   // ---- epilogue: D[r] -> row (r&3)+8*(r>>2)+4*(lane>>5), col lane&31
-  const long long nbase = n0 + wid * 32 + (lane & 31);
-  if (nbase < Np) {
+  #pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    const long long nbase = n0 + wid * 64 + half * 32 + (lane & 31);
+    if (nbase >= Np) continue;
+    const f32x16& acc = half ? acc1 : acc0;
     #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
