@@ -77,15 +77,21 @@ class FusedEngine:
         self.buffers = {k: v.detach().clone() for k, v in self.base.named_buffers()}
         self.client_chunk = client_chunk
         self._grad_fn = None
-        # population path: direct MFMA popconv kernels for supported models
-        # (GPU; CPU uses it only when forced, for parity tests)
+        # population path: direct MFMA popconv kernels for supported models.
+        # Measured round 1: the popconv round runs ~467 ms vs 229 ms for the
+        # vmap+hipGraph path on the headline config (popconv_fwd reaches
+        # 57-60 TF f32 = 37% of MFMA peak, but padded-plane FLOP inflation
+        # on the deep layers plus the unfused BN/elementwise chain eat the
+        # win) — so it is OPT-IN (BLADES_AMD_POPCONV=1 / FORCE_POP for CPU
+        # parity tests) until the kernel is pipelined and BN is fused.
         from blades_amd.engine.popmodel import population_forward_for
 
         self._pop_fn = None
-        if os.environ.get("BLADES_AMD_NO_POPCONV", "0") != "1":
+        want_pop = (os.environ.get("BLADES_AMD_POPCONV") == "1"
+                    or os.environ.get("BLADES_AMD_FORCE_POP") == "1")
+        if want_pop and os.environ.get("BLADES_AMD_NO_POPCONV", "0") != "1":
             fn = population_forward_for(model)
-            if fn is not None and (self.device.type == "cuda"
-                                   or os.environ.get("BLADES_AMD_FORCE_POP") == "1"):
+            if fn is not None:
                 self._pop_fn = fn
 
     # ------------------------------------------------------------ internals

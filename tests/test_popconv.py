@@ -149,7 +149,7 @@ def test_pop_engine_gpu_matches_vmap():
     from blades_amd.datasets import SyntheticFLDataset
 
     def run(no_pop):
-        env = dict(BLADES_AMD_NO_POPCONV="1") if no_pop else {}
+        env = {} if no_pop else dict(BLADES_AMD_POPCONV="1")
         old = {k: _os.environ.get(k) for k in env}
         _os.environ.update(env)
         try:
