@@ -20,6 +20,9 @@ from .centeredclipping import Centeredclipping
 from .clustering import Clustering
 from .clippedclustering import Clippedclustering
 from .fltrust import Fltrust
+from .byzantinesgd import Byzantinesgd
+from .async_ import (_AsyncCenteredClipping, _AsyncMean,
+                     _BaseAsyncAggregator, _DecentralizedAggregator)
 
 _REGISTRY = {
     "mean": Mean,
@@ -33,6 +36,7 @@ _REGISTRY = {
     "clustering": Clustering,
     "clippedclustering": Clippedclustering,
     "fltrust": Fltrust,
+    "byzantinesgd": Byzantinesgd,
 }
 
 
@@ -54,5 +58,8 @@ def register_aggregator(name: str):
 __all__ = [
     "Mean", "Median", "Trimmedmean", "Krum", "Multikrum", "Geomed", "Autogm",
     "Centeredclipping", "Clustering", "Clippedclustering", "Fltrust",
+    "Byzantinesgd",
     "get_aggregator", "register_aggregator", "_BaseAggregator",
+    "_BaseAsyncAggregator", "_AsyncMean", "_AsyncCenteredClipping",
+    "_DecentralizedAggregator",
 ]

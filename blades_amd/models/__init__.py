@@ -11,7 +11,7 @@ from torch import nn
 
 from .mlp import MLP
 from .resnet import ResNet, WideResNet, resnet18, wide_resnet28_10
-from .cct import CCT, CCTNet, cct_2_3x2_32
+from .cct import CCT, CCTNet, cct_2_3x2_32, cvt_2_4_32, vit_lite_2_4_32
 
 _REGISTRY = {
     "mlp": lambda **kw: MLP(**kw),
@@ -20,6 +20,8 @@ _REGISTRY = {
     "wideresnet28_10": lambda **kw: wide_resnet28_10(**kw),
     "cct": lambda **kw: cct_2_3x2_32(**kw),
     "cct_2_3x2_32": lambda **kw: cct_2_3x2_32(**kw),
+    "cvt": lambda **kw: cvt_2_4_32(**kw),
+    "vit_lite": lambda **kw: vit_lite_2_4_32(**kw),
 }
 
 
@@ -44,6 +46,7 @@ def num_params(model: nn.Module) -> int:
 
 __all__ = [
     "MLP", "ResNet", "WideResNet", "CCT", "CCTNet",
-    "resnet18", "wide_resnet28_10", "cct_2_3x2_32",
+    "resnet18", "wide_resnet28_10", "cct_2_3x2_32", "cvt_2_4_32",
+    "vit_lite_2_4_32",
     "get_model", "register_model", "num_params",
 ]

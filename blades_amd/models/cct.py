@@ -107,18 +107,54 @@ class EncoderLayer(nn.Module):
         return x + self.drop_path(self.dropout2(h))
 
 
+class PatchEmbed(nn.Module):
+    """Non-overlapping patch embedding (ViT-Lite / CVT tokenizer)."""
+
+    def __init__(self, in_channels: int = 3, embedding_dim: int = 128,
+                 patch_size: int = 4):
+        super().__init__()
+        self.proj = nn.Conv2d(in_channels, embedding_dim, patch_size,
+                              stride=patch_size)
+
+    def sequence_length(self, in_channels=3, height=32, width=32) -> int:
+        with torch.no_grad():
+            return self.forward(torch.zeros(1, in_channels, height, width)).shape[1]
+
+    def forward(self, x):
+        return self.proj(x).flatten(2, 3).transpose(-2, -1)
+
+
 class CCT(nn.Module):
+    """Compact transformer family.
+
+    ``tokenizer='conv'`` + ``pool='seq'``  -> CCT (default);
+    ``tokenizer='patch'`` + ``pool='seq'`` -> CVT;
+    ``tokenizer='patch'`` + ``pool='cls'`` -> ViT-Lite.
+    (reference zoo: cctnets/cct.py, cvt.py, vit.py)
+    """
+
     def __init__(self, img_size: int = 32, embedding_dim: int = 128,
                  num_layers: int = 2, num_heads: int = 2, mlp_ratio: float = 1.0,
                  n_conv_layers: int = 2, num_classes: int = 10,
                  dropout: float = 0.0, attn_dropout: float = 0.1,
                  stochastic_depth: float = 0.1, in_channels: int = 3,
-                 positional_embedding: str = "learnable"):
+                 positional_embedding: str = "learnable",
+                 tokenizer: str = "conv", pool: str = "seq",
+                 patch_size: int = 4):
         super().__init__()
-        self.tokenizer = Tokenizer(n_conv_layers=n_conv_layers,
-                                   in_channels=in_channels,
-                                   embedding_dim=embedding_dim)
+        if tokenizer == "conv":
+            self.tokenizer = Tokenizer(n_conv_layers=n_conv_layers,
+                                       in_channels=in_channels,
+                                       embedding_dim=embedding_dim)
+        else:
+            self.tokenizer = PatchEmbed(in_channels=in_channels,
+                                        embedding_dim=embedding_dim,
+                                        patch_size=patch_size)
+        self.pool = pool
         seq_len = self.tokenizer.sequence_length(in_channels, img_size, img_size)
+        if pool == "cls":
+            self.class_emb = nn.Parameter(torch.zeros(1, 1, embedding_dim))
+            seq_len += 1
         self.seq_len = seq_len
 
         if positional_embedding == "learnable":
@@ -162,21 +198,41 @@ class CCT(nn.Module):
 
     def forward(self, x):
         x = self.tokenizer(x)
+        if self.pool == "cls":
+            cls = self.class_emb.expand(x.shape[0], -1, -1)
+            x = torch.cat([cls, x], dim=1)
         if self.positional_emb is not None:
             x = x + self.positional_emb
         x = self.dropout(x)
         for blk in self.blocks:
             x = blk(x)
         x = self.norm(x)
-        # sequence pooling: softmax(Wx) over tokens, weighted sum
-        w = F.softmax(self.attention_pool(x), dim=1)  # [B, N, 1]
-        x = torch.matmul(w.transpose(-1, -2), x).squeeze(-2)
+        if self.pool == "cls":
+            x = x[:, 0]
+        else:
+            # sequence pooling: softmax(Wx) over tokens, weighted sum
+            w = F.softmax(self.attention_pool(x), dim=1)  # [B, N, 1]
+            x = torch.matmul(w.transpose(-1, -2), x).squeeze(-2)
         return self.fc(x)
 
 
 def cct_2_3x2_32(num_classes: int = 10, img_size: int = 32, **kw) -> CCT:
     return CCT(img_size=img_size, embedding_dim=128, num_layers=2, num_heads=2,
                mlp_ratio=1.0, n_conv_layers=2, num_classes=num_classes, **kw)
+
+
+def cvt_2_4_32(num_classes: int = 10, img_size: int = 32, **kw) -> CCT:
+    """Compact Vision Transformer: patch embedding + sequence pooling."""
+    return CCT(img_size=img_size, embedding_dim=128, num_layers=2, num_heads=2,
+               mlp_ratio=1.0, num_classes=num_classes, tokenizer="patch",
+               pool="seq", patch_size=4, **kw)
+
+
+def vit_lite_2_4_32(num_classes: int = 10, img_size: int = 32, **kw) -> CCT:
+    """ViT-Lite: patch embedding + class token."""
+    return CCT(img_size=img_size, embedding_dim=128, num_layers=2, num_heads=2,
+               mlp_ratio=1.0, num_classes=num_classes, tokenizer="patch",
+               pool="cls", patch_size=4, **kw)
 
 
 class CCTNet(nn.Module):

@@ -1,0 +1,118 @@
+"""Tests for the long-tail components: ByzantineSGD, async/decentralized
+aggregation, torch_utils shims, LEAF helpers, model-zoo extras, algorithms."""
+import numpy as np
+import pytest
+import torch
+
+from blades_amd.aggregators import (Byzantinesgd, _AsyncCenteredClipping,
+                                    _AsyncMean, _DecentralizedAggregator,
+                                    get_aggregator)
+from blades_amd.aggregators.torch_utils import (HLoss, clip_para_norm_,
+                                                clip_tensor_norm_, cos_sim,
+                                                l2dist, l2norm)
+from blades_amd.algorithms import fedavg, fedsgd
+from blades_amd.datasets.leaf import (iid_divide, remove_small_users,
+                                      sample_iid, sample_noniid,
+                                      train_test_split)
+from blades_amd.models import cvt_2_4_32, get_model, vit_lite_2_4_32
+
+
+def test_byzantinesgd_filters_outlier():
+    g = torch.Generator().manual_seed(0)
+    U = torch.randn(8, 16, generator=g) * 0.1
+    U[0] = 100.0  # gross outlier
+    agg = Byzantinesgd(num_clients=8, th_A=5.0, th_B=5.0, th_V=5.0)
+    out = agg(list(U))
+    assert 0 not in agg.good
+    assert torch.allclose(out, U[agg.good].mean(0))
+    # state round-trips
+    st = agg.state_dict()
+    agg2 = Byzantinesgd(num_clients=8, th_A=5.0, th_B=5.0, th_V=5.0)
+    agg2.load_state_dict(st)
+    assert agg2.good == agg.good
+    assert get_aggregator("byzantinesgd", num_clients=4) is not None
+
+
+def test_async_mean_normalizes_by_population():
+    a = torch.ones(4)
+    out = _AsyncMean()([a, None, a, None])
+    assert torch.allclose(out, torch.full((4,), 0.5))
+
+
+def test_async_centeredclipping_runs():
+    g = torch.Generator().manual_seed(1)
+    U = list(torch.randn(5, 8, generator=g))
+    agg = _AsyncCenteredClipping(tau=2.0, n_iter=3)
+    out1 = agg([U[0], None, U[2], U[3], None])
+    out2 = agg(U)
+    assert out1.shape == out2.shape == (8,)
+    assert not torch.allclose(out1, out2)  # momentum advanced
+
+
+def test_decentralized_gossip_row():
+    w = torch.tensor([0.5, 0.25, 0.25])
+    agg = _DecentralizedAggregator(node_index=0, neighbor_indices=[1, 2],
+                                   weights=w)
+    xs = [torch.ones(3), torch.full((3,), 2.0), torch.full((3,), 4.0)]
+    out = agg(xs)
+    assert torch.allclose(out, torch.full((3,), 0.5 + 0.5 + 1.0))
+
+
+def test_clip_tensor_norm():
+    v = torch.full((10,), 3.0)
+    clip_tensor_norm_(v, max_norm=1.0)
+    assert abs(v.norm().item() - 1.0) < 1e-4
+    # under the norm: untouched
+    u = torch.full((4,), 0.1)
+    clip_tensor_norm_(u, max_norm=10.0)
+    assert torch.allclose(u, torch.full((4,), 0.1))
+
+
+def test_clip_para_norm_and_model_dict_helpers():
+    d1 = {"a": torch.ones(4), "b": torch.ones(2)}
+    total = clip_para_norm_(dict(d1), max_norm=1.0)
+    assert total > 1.0
+    d2 = {"a": torch.ones(4), "b": torch.ones(2)}
+    assert l2dist(d2, d2) == 0
+    assert l2norm(d2) > 0
+    assert abs(cos_sim(d2, d2).item() - 1.0) < 1e-4
+    h = HLoss()(torch.randn(3, 5))
+    assert torch.isfinite(h)
+
+
+def test_leaf_helpers():
+    rng = np.random.RandomState(0)
+    y = rng.randint(0, 10, 200)
+    groups = iid_divide(list(range(100)), 7)
+    assert sum(len(g) for g in groups) == 100
+    iid = sample_iid(y, 8, rng)
+    assert sum(len(v) for v in iid.values()) == 200
+    noniid = sample_noniid(y, 10, shards_per_user=2, rng=rng)
+    # shard sampling gives each user few distinct labels
+    distinct = [len(np.unique(y[v])) for v in noniid.values()]
+    assert np.mean(distinct) < 6
+    tr, te = train_test_split(np.arange(50), frac=0.8, rng=rng)
+    assert len(tr) == 40 and len(te) == 10
+    kept = remove_small_users({0: np.arange(3), 1: np.arange(30)}, 10)
+    assert list(kept) == [1]
+
+
+def test_model_zoo_extras():
+    for fn in (cvt_2_4_32, vit_lite_2_4_32):
+        m = fn()
+        assert m(torch.randn(2, 3, 32, 32)).shape == (2, 10)
+    assert get_model("cvt") is not None
+    assert get_model("vit_lite") is not None
+
+
+def test_algorithm_presets(tmp_path):
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import MLP
+
+    ds = SyntheticFLDataset(num_clients=4, samples_per_client=8, batch_size=4,
+                            shape=(1, 28, 28), seed=0)
+    sim = Simulator(ds, log_path=str(tmp_path), seed=0)
+    ret = sim.run(MLP(), global_rounds=1, validate_interval=0, **fedsgd())
+    assert len(ret) == 1
+    assert fedavg(local_steps=3)["local_steps"] == 3
