@@ -103,6 +103,46 @@ class DistributedRuntime:
         blocks = [out[r * kmax: r * kmax + sizes[r]] for r in range(self.world_size)]
         return torch.cat(blocks, dim=0)
 
+    def all_to_all_coordinate_shard(self, local: torch.Tensor,
+                                    total_rows: int,
+                                    dshard: int) -> torch.Tensor:
+        """Client-sharded [K/ws, ws*dshard] -> coordinate-sharded
+        [K, dshard] (this rank's coordinate slice of EVERY client's update).
+
+        The SP-style re-shard of SURVEY.md §5.7: one all-to-all moves
+        (K/ws)·d·(ws-1)/ws bytes per rank — ws× less traffic than a full
+        all-gather — and aggregation work splits d/ws per rank.  Rows are
+        padded to the max shard size (all_to_all_single needs equal splits)
+        and re-ordered back to global client order.
+        """
+        if not self.distributed:
+            return local
+        sizes = [len(s) for s in self.shard_indices(total_rows)]
+        kmax = max(sizes)
+        ws = self.world_size
+        if local.shape[0] < kmax:
+            pad = torch.zeros(kmax - local.shape[0], local.shape[1],
+                              device=local.device, dtype=local.dtype)
+            send = torch.cat([local, pad], dim=0)
+        else:
+            send = local
+        # [kmax, ws, dshard] -> [ws, kmax, dshard] so split s goes to rank s
+        send = send.view(kmax, ws, dshard).transpose(0, 1).contiguous()
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send)
+        # recv[r] = rank r's rows (padded), MY coordinate slice
+        blocks = [recv[r, :sizes[r]] for r in range(ws)]
+        return torch.cat(blocks, dim=0)  # [K, dshard], global client order
+
+    def all_gather_flat(self, shard: torch.Tensor) -> torch.Tensor:
+        """Gather equal-size 1-D shards into one vector (rank order)."""
+        if not self.distributed:
+            return shard
+        out = torch.empty(self.world_size * shard.numel(),
+                          device=shard.device, dtype=shard.dtype)
+        dist.all_gather_into_tensor(out, shard.contiguous())
+        return out
+
     def all_reduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         if self.distributed:
             dist.all_reduce(t, op=dist.ReduceOp.SUM if op == "sum" else dist.ReduceOp.MAX)

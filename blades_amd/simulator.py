@@ -88,6 +88,7 @@ class Simulator:
         client_chunk: Optional[int] = None,
         device: Optional[str] = None,
         hip_graphs: bool = True,
+        gather: str = "auto",
         **kwargs,
     ):
         self.use_actor = mode == "actor"
@@ -159,6 +160,14 @@ class Simulator:
         self._graph_round = None
         self._theta: Optional[torch.Tensor] = None
         self._model_stale = False  # True while θ (graph mode) is ahead of model
+        # update-gather strategy: "full" all-gathers the whole U on every
+        # rank; "coordinate" re-shards by coordinate (SP-style all-to-all,
+        # SURVEY.md §5.7) — mandatory at 1e4-client × WRN scale where U is
+        # 1.46 TB; "auto" picks coordinate when the aggregator/attacks
+        # support it
+        if gather not in ("auto", "full", "coordinate"):
+            raise ValueError(f"gather must be auto|full|coordinate, got {gather}")
+        self._gather = gather
 
     # ------------------------------------------------------------ builders
     def _init_aggregator(self, aggregator, aggregator_kws) -> None:
@@ -292,6 +301,36 @@ class Simulator:
         self._model_stale = True
         return True
 
+    def _coordinate_supported(self) -> Optional[str]:
+        """Reason the coordinate-sharded gather can NOT be used, else None."""
+        if not self.runtime.distributed:
+            return "single rank (nothing to re-shard)"
+        if not getattr(self.aggregator, "coordinate_shardable", False):
+            return f"aggregator {type(self.aggregator).__name__} needs full rows"
+        from blades_amd.attackers import (AlieClient, IpmClient,
+                                          LabelflippingClient, NoiseClient,
+                                          SignflippingClient)
+        from blades_amd.client import uses_default_training
+
+        ok_types = (AlieClient, IpmClient, NoiseClient, LabelflippingClient,
+                    SignflippingClient)
+        for c in self.get_clients():
+            if c.is_byzantine() and type(c) not in ok_types:
+                return f"custom byzantine client {type(c).__name__}"
+            if not uses_default_training(c):
+                return f"custom client {type(c).__name__}"
+        return None
+
+    def _use_coordinate(self) -> bool:
+        if self._gather == "full":
+            return False
+        reason = self._coordinate_supported()
+        if reason is None:
+            return True
+        if self._gather == "coordinate":
+            raise RuntimeError(f"gather='coordinate' not possible: {reason}")
+        return False
+
     def train_round(self, global_round: int, local_steps: int,
                     clients: List[BladesClient], lr: float) -> None:
         """One global round (reference: train_actor, simulator.py:203-247)."""
@@ -302,11 +341,20 @@ class Simulator:
         all_clients = self.get_clients()
         rows = {c.id(): i for i, c in enumerate(all_clients)}
         shard = rt.my_shard(all_clients)
+        coordinate = self._use_coordinate()
 
         # slab rows padded to a float4 multiple: the HIP kernels take the
-        # 16B-vectorized path on every row while views stay zero-copy
+        # 16B-vectorized path on every row while views stay zero-copy.
+        # Coordinate mode additionally pads to a world-size multiple so the
+        # all-to-all splits evenly.
         d = self._spec.d
-        d_pad = -(-d // 4) * 4
+        if coordinate:
+            per_rank = -(-d // rt.world_size)          # ceil(d / ws)
+            dshard = -(-per_rank // 4) * 4             # round up to float4
+            d_pad = dshard * rt.world_size
+        else:
+            dshard = 0
+            d_pad = -(-d // 4) * 4
         with trace_range("blades/local_train"):
             theta = self.server.flat_parameters(device=self.device,
                                                 out=self._theta)
@@ -332,6 +380,11 @@ class Simulator:
                     for c in custom:
                         U_local[local_pos[c.id()]].copy_(
                             updates[c.id()].to(self.device))
+
+        if coordinate:
+            self._aggregate_coordinate(global_round, buf_local, U_local,
+                                       all_clients, shard, rows, dshard, d)
+            return
 
         with trace_range("blades/gather"):
             buf = rt.all_gather_rows(buf_local, total_rows=len(all_clients))
@@ -369,6 +422,73 @@ class Simulator:
 
         with trace_range("blades/apply"):
             self.server.apply_update(aggregated.to(self.device))
+
+    def _aggregate_coordinate(self, global_round: int, buf_local, U_local,
+                              all_clients, shard, rows, dshard: int,
+                              d: int) -> None:
+        """SP-style aggregation: all-to-all to coordinate shards [K, d/ws],
+        attacks + aggregation on the shard, all-gather of the Δ shards.
+
+        The only full-row state that ever exists is the rank-local client
+        shard (clients in my shard keep their update views; cross-rank rows
+        are never materialized — this is what makes the 1e4-client × WRN
+        configs fit in 288 GB/GPU, SURVEY.md §7 hard-part 2).
+        """
+        from blades_amd.attackers import AlieClient, IpmClient, NoiseClient
+        from blades_amd.ops import philox_normal
+        from blades_amd.utils import client_philox_seed
+
+        rt = self.runtime
+        local_pos = {c.id(): i for i, c in enumerate(shard)}
+
+        # noise attackers need their FULL row -> craft pre-reshard on the
+        # owner rank (deterministic per (client, round): layout-invariant)
+        for c in shard:
+            if type(c) is NoiseClient:
+                seed = client_philox_seed(self._seed, rows[c.id()],
+                                          global_round, tag=12)
+                U_local[local_pos[c.id()]].copy_(philox_normal(
+                    (d,), c._noise_mean, c._noise_std, seed,
+                    device=self.device))
+            c.save_update_view(U_local[local_pos[c.id()]])
+
+        with trace_range("blades/reshard"):
+            Ucoord = rt.all_to_all_coordinate_shard(
+                buf_local, total_rows=len(all_clients), dshard=dshard)
+            torch.nan_to_num_(Ucoord)
+
+        with trace_range("blades/attack"):
+            honest = torch.tensor([not c.is_byzantine() for c in all_clients],
+                                  device=self.device)
+            n_honest = int(honest.sum())
+            from blades_amd import ops as _ops
+
+            alie_groups: Dict[float, List[int]] = {}
+            ipm_groups: Dict[float, List[int]] = {}
+            for i, c in enumerate(all_clients):
+                if type(c) is AlieClient:
+                    alie_groups.setdefault(float(c.z_max), []).append(i)
+                elif type(c) is IpmClient:
+                    ipm_groups.setdefault(float(c.epsilon), []).append(i)
+            for z, rws in alie_groups.items():
+                mu, std = _ops.masked_col_mean_std(Ucoord, honest,
+                                                   unbiased=True,
+                                                   count=n_honest)
+                idx = torch.tensor(rws, device=self.device)
+                Ucoord.index_copy_(0, idx, (mu - std * z).unsqueeze(0)
+                                   .expand(len(rws), -1))
+            for eps, rws in ipm_groups.items():
+                hm = _ops.masked_col_mean(Ucoord, honest, count=n_honest)
+                idx = torch.tensor(rws, device=self.device)
+                Ucoord.index_copy_(0, idx, (-eps * hm).unsqueeze(0)
+                                   .expand(len(rws), -1))
+
+        with trace_range("blades/aggregate"):
+            delta_shard = self.aggregator(Ucoord)
+
+        with trace_range("blades/apply"):
+            delta = rt.all_gather_flat(delta_shard)[:d]
+            self.server.apply_update(delta)
 
     # reference-name alias
     def train_actor(self, global_round: int, num_rounds: int,

@@ -81,3 +81,114 @@ def test_world2_matches_single_rank():
     assert l0 == pytest.approx(l1)
     # and they match the single-rank run exactly (layout-invariant RNG)
     assert torch.allclose(t0, theta1, atol=1e-6)
+
+
+def _coord_worker(rank, world_size, port, out_q, gather, seed=9):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world_size),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    ds = SyntheticFLDataset(num_clients=8, samples_per_client=16, batch_size=8,
+                            shape=(1, 28, 28), num_classes=10, seed=0)
+    sim = Simulator(ds, num_byzantine=3, attack="alie",
+                    attack_kws={"num_clients": 8, "num_byzantine": 3},
+                    aggregator="median", log_path=f"/tmp/bl_coord_{rank}",
+                    seed=seed, gather=gather)
+    torch.manual_seed(seed)
+    sim.run(MLP(), global_rounds=3, validate_interval=0, client_lr=0.1)
+    out_q.put((rank, sim.server.flat_parameters().numpy().copy()))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_coordinate_shard_matches_full():
+    """SP-style coordinate-sharded aggregation == full-gather == single
+    rank, bit-for-bit (layout-invariant RNG, deterministic CPU math)."""
+    # single-rank baseline
+    import numpy as np
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    ds = SyntheticFLDataset(num_clients=8, samples_per_client=16, batch_size=8,
+                            shape=(1, 28, 28), num_classes=10, seed=0)
+    sim = Simulator(ds, num_byzantine=3, attack="alie",
+                    attack_kws={"num_clients": 8, "num_byzantine": 3},
+                    aggregator="median", log_path="/tmp/bl_coord_single",
+                    seed=9)
+    torch.manual_seed(9)
+    sim.run(MLP(), global_rounds=3, validate_interval=0, client_lr=0.1)
+    theta1 = sim.server.flat_parameters().numpy()
+
+    for gather in ("coordinate", "full"):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        port = 29621 if gather == "coordinate" else 29622
+        procs = [ctx.Process(target=_coord_worker,
+                             args=(r, WORLD, port, q, gather))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(WORLD):
+            rank, theta = q.get(timeout=240)
+            results[rank] = theta
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        assert np.array_equal(results[0], results[1]), gather
+        assert np.allclose(results[0], theta1, atol=1e-6), gather
+
+
+@pytest.mark.timeout(300)
+def test_coordinate_with_noise_attack_matches_full():
+    """Noise attackers craft rows pre-reshard; results must still match the
+    full-gather path exactly."""
+    import numpy as np
+
+    def worker(rank, world_size, port, out_q, gather):
+        os.environ.update({
+            "RANK": str(rank), "WORLD_SIZE": str(world_size),
+            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        })
+        from blades_amd import Simulator
+        from blades_amd.datasets import SyntheticFLDataset
+
+        ds = SyntheticFLDataset(num_clients=6, samples_per_client=16,
+                                batch_size=8, shape=(1, 28, 28),
+                                num_classes=10, seed=0)
+        sim = Simulator(ds, num_byzantine=2, attack="noise",
+                        aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                        log_path=f"/tmp/bl_cnoise_{gather}_{rank}", seed=4,
+                        gather=gather)
+        torch.manual_seed(4)
+        sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
+        out_q.put((rank, sim.server.flat_parameters().numpy().copy()))
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+    outs = {}
+    for gather in ("coordinate", "full"):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        port = 29631 if gather == "coordinate" else 29632
+        procs = [ctx.Process(target=worker, args=(r, WORLD, port, q, gather))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        res = {}
+        for _ in range(WORLD):
+            rank, theta = q.get(timeout=240)
+            res[rank] = theta
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        outs[gather] = res[0]
+    # trimmed-mean sums reduce in a layout-dependent order (sharded columns
+    # are strided views) -> occasional 1-ulp differences
+    assert np.allclose(outs["coordinate"], outs["full"], atol=1e-7)
