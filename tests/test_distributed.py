@@ -144,40 +144,41 @@ def test_coordinate_shard_matches_full():
         assert np.allclose(results[0], theta1, atol=1e-6), gather
 
 
+def _noise_worker(rank, world_size, port, out_q, gather):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world_size),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    ds = SyntheticFLDataset(num_clients=6, samples_per_client=16,
+                            batch_size=8, shape=(1, 28, 28),
+                            num_classes=10, seed=0)
+    sim = Simulator(ds, num_byzantine=2, attack="noise",
+                    aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                    log_path=f"/tmp/bl_cnoise_{gather}_{rank}", seed=4,
+                    gather=gather)
+    torch.manual_seed(4)
+    sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
+    out_q.put((rank, sim.server.flat_parameters().numpy().copy()))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
 @pytest.mark.timeout(300)
 def test_coordinate_with_noise_attack_matches_full():
     """Noise attackers craft rows pre-reshard; results must still match the
     full-gather path exactly."""
     import numpy as np
 
-    def worker(rank, world_size, port, out_q, gather):
-        os.environ.update({
-            "RANK": str(rank), "WORLD_SIZE": str(world_size),
-            "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
-            "MASTER_PORT": str(port),
-        })
-        from blades_amd import Simulator
-        from blades_amd.datasets import SyntheticFLDataset
-
-        ds = SyntheticFLDataset(num_clients=6, samples_per_client=16,
-                                batch_size=8, shape=(1, 28, 28),
-                                num_classes=10, seed=0)
-        sim = Simulator(ds, num_byzantine=2, attack="noise",
-                        aggregator="trimmedmean", aggregator_kws={"nb": 2},
-                        log_path=f"/tmp/bl_cnoise_{gather}_{rank}", seed=4,
-                        gather=gather)
-        torch.manual_seed(4)
-        sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
-        out_q.put((rank, sim.server.flat_parameters().numpy().copy()))
-        import torch.distributed as dist
-        dist.destroy_process_group()
-
     outs = {}
     for gather in ("coordinate", "full"):
         ctx = mp.get_context("spawn")
         q = ctx.Queue()
         port = 29631 if gather == "coordinate" else 29632
-        procs = [ctx.Process(target=worker, args=(r, WORLD, port, q, gather))
+        procs = [ctx.Process(target=_noise_worker, args=(r, WORLD, port, q, gather))
                  for r in range(WORLD)]
         for p in procs:
             p.start()
