@@ -203,6 +203,148 @@ __global__ void trimmed_select_kernel(const float* __restrict__ U,
 }
 
 // ---------------------------------------------------------------------------
+// K2/K3 at large K — dual radix-select trimmed mean
+// ---------------------------------------------------------------------------
+// The LDS streaming-selection kernel above needs 2·b floats of LDS per
+// thread, capping b ≈ 127.  Config 3/5 scales (K = 1e3..1e4, median ⇒
+// b = (K−1)/2) use this kernel instead: per coordinate, find BOTH trim
+// thresholds (the b-th and (K−b−1)-th order statistics) by 4 levels of
+// byte-wise MSB radix selection on the order-preserving fp32→u32 key, then
+// one final pass sums the kept middle band with exact tie handling.
+// 5 passes over the slab total; each block owns a 32-coordinate tile with
+// two [32][256] u32 histograms in LDS (atomic bin increments), per-tile
+// element reads fully coalesced (consecutive lanes → consecutive columns).
+
+__device__ __forceinline__ unsigned int order_key(float x) {
+  unsigned int u = __float_as_uint(x);
+  return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+}
+
+__device__ __forceinline__ float key_value(unsigned int key) {
+  unsigned int u = (key & 0x80000000u) ? (key & 0x7fffffffu) : ~key;
+  return __uint_as_float(u);
+}
+
+constexpr int RT_T = 32;  // coordinates per block tile
+
+__global__ __launch_bounds__(256)
+void radix_trimmed_kernel(const float* __restrict__ U,
+                          float* __restrict__ out,
+                          long long K, long long d, long long ld, long b) {
+  __shared__ unsigned int histA[RT_T][256];
+  __shared__ unsigned int histB[RT_T][256];
+  __shared__ unsigned int prefA[RT_T], baseA[RT_T];
+  __shared__ unsigned int prefB[RT_T], baseB[RT_T];
+  __shared__ double partial[RT_T][8];      // final-pass per-coord partials
+  __shared__ unsigned int eqA[RT_T][8], eqB[RT_T][8];
+
+  const long long kA = b;             // first kept rank (0-indexed)
+  const long long kB = K - b - 1;     // last kept rank
+
+  for (long long j0 = (long long)blockIdx.x * RT_T; j0 < d;
+       j0 += (long long)gridDim.x * RT_T) {
+    const int tw = (int)((d - j0) < RT_T ? (d - j0) : RT_T);
+    if (threadIdx.x < RT_T) {
+      prefA[threadIdx.x] = 0; baseA[threadIdx.x] = 0;
+      prefB[threadIdx.x] = 0; baseB[threadIdx.x] = 0;
+    }
+    for (int level = 3; level >= 0; --level) {
+      __syncthreads();
+      for (int e = threadIdx.x; e < RT_T * 256; e += blockDim.x) {
+        (&histA[0][0])[e] = 0;
+        (&histB[0][0])[e] = 0;
+      }
+      __syncthreads();
+      const int shift = 8 * level;
+      for (long long e = threadIdx.x; e < K * RT_T; e += blockDim.x) {
+        const int t = (int)(e % RT_T);
+        if (t >= tw) continue;
+        const long long k = e / RT_T;
+        const unsigned int key = order_key(U[k * ld + j0 + t]);
+        // bytes above this level must match the already-selected prefix
+        const unsigned int hi = (level == 3) ? 0u : (key >> (shift + 8));
+        const unsigned int bin = (key >> shift) & 255u;
+        if (hi == prefA[t]) atomicAdd(&histA[t][bin], 1u);
+        if (hi == prefB[t]) atomicAdd(&histB[t][bin], 1u);
+      }
+      __syncthreads();
+      if (threadIdx.x < (unsigned)tw) {
+        const int t = threadIdx.x;
+        unsigned long long cum = baseA[t];
+        for (int bin = 0; bin < 256; ++bin) {
+          const unsigned int c = histA[t][bin];
+          if (cum + c > (unsigned long long)kA) {
+            prefA[t] = (prefA[t] << 8) | (unsigned)bin;
+            baseA[t] = (unsigned)cum;
+            break;
+          }
+          cum += c;
+        }
+        cum = baseB[t];
+        for (int bin = 0; bin < 256; ++bin) {
+          const unsigned int c = histB[t][bin];
+          if (cum + c > (unsigned long long)kB) {
+            prefB[t] = (prefB[t] << 8) | (unsigned)bin;
+            baseB[t] = (unsigned)cum;
+            break;
+          }
+          cum += c;
+        }
+      }
+    }
+    __syncthreads();
+    // final pass: 8 threads per coordinate accumulate the strict-interior
+    // sum and the tie counts at both thresholds
+    {
+      const int t = threadIdx.x & (RT_T - 1);
+      const int sl = threadIdx.x / RT_T;  // 0..7
+      double s = 0.0;
+      unsigned int ea = 0, eb = 0;
+      if (t < tw) {
+        const unsigned int keyA = prefA[t], keyB = prefB[t];
+        for (long long k = sl; k < K; k += 8) {
+          const float x = U[k * ld + j0 + t];
+          const unsigned int key = order_key(x);
+          if (key > keyA && key < keyB) s += x;
+          ea += (key == keyA);
+          eb += (key == keyB);
+        }
+      }
+      partial[t][sl] = s;
+      eqA[t][sl] = ea;
+      eqB[t][sl] = eb;
+    }
+    __syncthreads();
+    if (threadIdx.x < (unsigned)tw) {
+      const int t = threadIdx.x;
+      double s = 0.0;
+      long long ea = 0, eb = 0;
+      for (int i = 0; i < 8; ++i) {
+        s += partial[t][i];
+        ea += eqA[t][i];
+        eb += eqB[t][i];
+      }
+      const double vA = (double)key_value(prefA[t]);
+      const double vB = (double)key_value(prefB[t]);
+      const long long cltA = baseA[t], cltB = baseB[t];
+      double total;
+      if (prefA[t] == prefB[t]) {
+        total = (double)(K - 2 * b) * vA;
+      } else {
+        // kept ranks are [b, K-b); ties at the thresholds contribute the
+        // overlap of their rank range with the kept band
+        const long long incA =
+            (cltA + ea < K - b ? cltA + ea : K - b) - b;
+        const long long incB = (K - b) - (cltB > b ? cltB : b);
+        total = s + (double)incA * vA + (double)incB * vB;
+      }
+      out[j0 + t] = (float)(total / (double)(K - 2 * b));
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Row reductions (K6 Weiszfeld distances, K7 clip norms, K9 FLTrust dots)
 // ---------------------------------------------------------------------------
 // grid = (splits, K): block (s, k) reduces slice s of row k with float4
@@ -466,11 +608,19 @@ torch::Tensor trimmed_mean_select(torch::Tensor U, long b) {
   int BS = 256;
   size_t need = 2 * (size_t)b * 4 * BS;
   while (BS > 64 && need > 64 * 1024) { BS /= 2; need /= 2; }
-  TORCH_CHECK(need <= 64 * 1024,
-              "b too large for the LDS selection kernel (use torch fallback)");
-  const int grid = col_grid(v.d, BS);
-  trimmed_select_kernel<<<grid, BS, need, stream>>>(
-      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b);
+  if (need <= 64 * 1024 && b > 0) {
+    const int grid = col_grid(v.d, BS);
+    trimmed_select_kernel<<<grid, BS, need, stream>>>(
+        v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b);
+    return out;
+  }
+  if (b == 0) {
+    return col_mean(U) ;
+  }
+  // large b (median at K >= ~260): dual radix-select kernel
+  const int grid = (int)std::min<long long>(cdiv(v.d, RT_T), kMaxBlocks);
+  radix_trimmed_kernel<<<grid, 256, 0, stream>>>(
+      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b);
   return out;
 }
 

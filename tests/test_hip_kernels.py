@@ -160,3 +160,30 @@ def test_fail_loudly_without_force(ext, monkeypatch):
     with pytest.raises(RuntimeError, match="refusing to fall back"):
         ops.col_mean(U)
     monkeypatch.setattr(ops, "_EXT_ERR", None)
+
+
+@pytest.mark.parametrize("K,b", [(1000, 499), (1000, 300), (2000, 999),
+                                 (517, 258), (4096, 1000)])
+def test_radix_trimmed_large_K(ext, K, b):
+    """Dual radix-select path (large b beyond the LDS kernel's range)."""
+    from blades_amd.ops import torch_ref
+    U = randU(K, 20000, seed=K + b)
+    out = ext.trimmed_mean(U, b)
+    ref = torch_ref.trimmed_mean(U, b)
+    assert torch.allclose(out, ref, atol=1e-5, rtol=1e-5)
+
+
+def test_radix_trimmed_with_ties(ext):
+    from blades_amd.ops import torch_ref
+    U = randU(999, 4096, seed=1).round()  # heavy ties
+    out = ext.trimmed_mean(U, 400)
+    ref = torch_ref.trimmed_mean(U, 400)
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_col_median_large_K(ext):
+    from blades_amd.ops import torch_ref
+    for K in (999, 1000):
+        U = randU(K, 30000, seed=K)
+        assert torch.allclose(ext.col_median(U), torch_ref.col_median(U),
+                              atol=1e-6), K
