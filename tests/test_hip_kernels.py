@@ -165,12 +165,16 @@ def test_fail_loudly_without_force(ext, monkeypatch):
 @pytest.mark.parametrize("K,b", [(1000, 499), (1000, 300), (2000, 999),
                                  (517, 258), (4096, 1000)])
 def test_radix_trimmed_large_K(ext, K, b):
-    """Dual radix-select path (large b beyond the LDS kernel's range)."""
+    """Dual radix-select path (large b beyond the LDS kernel's range).
+
+    Reference computed in fp64: the torch fp32 sum-minus-topk form loses
+    ~1e-5 to cancellation when only 1-2 of K values survive the trim,
+    while the kernel accumulates in fp64."""
     from blades_amd.ops import torch_ref
     U = randU(K, 20000, seed=K + b)
     out = ext.trimmed_mean(U, b)
-    ref = torch_ref.trimmed_mean(U, b)
-    assert torch.allclose(out, ref, atol=1e-5, rtol=1e-5)
+    ref = torch_ref.trimmed_mean(U.double(), b).float()
+    assert torch.allclose(out, ref, atol=1e-6, rtol=1e-5)
 
 
 def test_radix_trimmed_with_ties(ext):
@@ -185,5 +189,5 @@ def test_col_median_large_K(ext):
     from blades_amd.ops import torch_ref
     for K in (999, 1000):
         U = randU(K, 30000, seed=K)
-        assert torch.allclose(ext.col_median(U), torch_ref.col_median(U),
-                              atol=1e-6), K
+        ref = torch_ref.col_median(U.double()).float()
+        assert torch.allclose(ext.col_median(U), ref, atol=1e-6), K
