@@ -26,8 +26,34 @@ import time
 import torch
 
 
+#: BASELINE.json named configs (per-GPU client counts; weak scaling)
+CONFIG_PRESETS = {
+    # 1: MNIST MLP FedSGD, 10 clients / 2 noise, Mean (CPU plumbing config)
+    "1": dict(model="mlp", clients_per_gpu=10, byz_frac=0.2, attack="noise",
+              aggregator="mean", local_steps=1, num_classes=10),
+    # 2: CIFAR10 ResNet-18 FedSGD, 100 clients / 20 ALIE, TrimmedMean (headline)
+    "2": dict(model="resnet18", clients_per_gpu=100, byz_frac=0.2,
+              attack="alie", aggregator="trimmedmean", local_steps=1,
+              num_classes=10),
+    # 3: CIFAR10 ResNet-18 FedAvg (5 local steps), 1000 clients, Median
+    "3": dict(model="resnet18", clients_per_gpu=125, byz_frac=0.0,
+              attack=None, aggregator="median", local_steps=5,
+              num_classes=10),
+    # 4: CIFAR10 ResNet-18, 100 clients / 20 IPM, Krum
+    "4": dict(model="resnet18", clients_per_gpu=100, byz_frac=0.2,
+              attack="ipm", aggregator="krum", local_steps=1,
+              num_classes=10),
+    # 5: CIFAR100 WRN-28-10 FedSGD, labelflipping, GeoMed (sharded scale)
+    "5": dict(model="wrn28_10", clients_per_gpu=32, byz_frac=0.1,
+              attack="labelflipping", aggregator="geomed", local_steps=1,
+              num_classes=100),
+}
+
+
 def parse_args():
     p = argparse.ArgumentParser()
+    p.add_argument("--config", type=str, default=None,
+                   help="BASELINE.json config preset 1-5 (overridable)")
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
@@ -41,7 +67,15 @@ def parse_args():
     p.add_argument("--client-chunk", type=int, default=None)
     p.add_argument("--num-classes", type=int, default=10)
     p.add_argument("--samples-per-client", type=int, default=64)
-    return p.parse_args()
+    args, _ = p.parse_known_args()
+    if args.config is not None:
+        preset = CONFIG_PRESETS[args.config]
+        defaults = {a.dest: a.default for a in p._actions}
+        for k, v in preset.items():
+            # presets fill only values the user left at their defaults
+            if getattr(args, k) == defaults.get(k):
+                setattr(args, k, v)
+    return args
 
 
 def main():
@@ -63,6 +97,9 @@ def main():
 
     total_clients = args.clients_per_gpu * n_gpus
     num_byz = int(round(args.byz_frac * total_clients))
+    if args.attack in (None, "none", "None"):
+        args.attack = None
+        num_byz = 0
 
     shape = (1, 28, 28) if args.model == "mlp" else (3, 32, 32)
     model_kw = {"num_classes": args.num_classes}
@@ -77,6 +114,8 @@ def main():
     attack_kws = {}
     if args.attack == "alie":
         attack_kws = {"num_clients": total_clients, "num_byzantine": num_byz}
+    elif args.attack == "labelflipping":
+        attack_kws = {"num_classes": args.num_classes}
 
     sim = Simulator(
         dataset=SyntheticFLDataset(
