@@ -322,7 +322,7 @@ class FusedEngine:
             test_set = dataset.get_all_test_data(c.id())
             loader = torch.utils.data.DataLoader(test_set, batch_size=batch_size)
             r = {"_meta": {"type": "client_validation"}, "E": round_number,
-                 "Length": 0, "Loss": 0.0}
+                 "Client": c.id(), "Length": 0, "Loss": 0.0}
             for name in metrics:
                 r[name] = 0.0
             for data, target in loader:

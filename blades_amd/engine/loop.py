@@ -44,5 +44,6 @@ class LoopEngine:
                 batch_size=batch_size,
                 metrics=metrics,
             )
+            r["Client"] = client.id()
             results.append(r)
         return results

@@ -268,6 +268,14 @@ class Simulator:
             client.device = self.device
             client.set_model(self.global_model, torch.optim.SGD, lr)
 
+    def get_phase_times(self) -> Dict[str, float]:
+        """Cumulative host-side seconds per round phase (local_train /
+        gather / reshard / attack / aggregate / apply / eval / graph_round)
+        — the rocprof-free first look at where a run spends time."""
+        from blades_amd.utils.tracing import phase_seconds
+
+        return dict(phase_seconds)
+
     def sync_model(self) -> None:
         """Write θ (graph-mode source of truth) back into the module params."""
         if self._model_stale:
@@ -519,6 +527,11 @@ class Simulator:
                     batch_size, self.metrics)
             gathered = rt.all_gather_object(local_metrics)
             metrics = [m for part in gathered for m in part]
+        # per-client validation records (reference: client.py:170 logged one
+        # per client from inside evaluate; here rank 0 logs the gathered set)
+        if rt.is_main():
+            for m in metrics:
+                self.json_logger.write(m)
         loss, top1 = self.log_validate(metrics)
         self.debug_logger.info(
             f"Test global round {global_round}, loss: {loss}, top1: {top1}")

@@ -62,10 +62,16 @@ def test_stats_log_is_json_lines(tmp_path):
     stats_file = os.path.join(str(tmp_path), "stats")
     lines = open(stats_file).read().strip().splitlines()
     assert len(lines) >= 2
+    types = set()
     for line in lines:
         rec = json.loads(line)
-        assert rec["_meta"]["type"] == "test"
-        assert {"Round", "top1", "Length", "Loss"} <= set(rec)
+        types.add(rec["_meta"]["type"])
+        if rec["_meta"]["type"] == "test":
+            assert {"Round", "top1", "Length", "Loss"} <= set(rec)
+        elif rec["_meta"]["type"] == "client_validation":
+            assert {"E", "Client", "Length", "Loss"} <= set(rec)
+    # both record kinds present (reference record taxonomy)
+    assert {"test", "client_validation"} <= types
     # the reference's consumer pattern also works on real JSON
     # (examples/Simulation on MNIST.py:69-81 replaces quotes then json.loads)
     json.loads(lines[0].replace("'", '"'))
