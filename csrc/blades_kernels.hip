@@ -628,6 +628,19 @@ torch::Tensor trimmed_mean(torch::Tensor U, long b) {
   return trimmed_mean_select(U, b);
 }
 
+torch::Tensor trimmed_mean_radix(torch::Tensor U, long b) {
+  // direct radix path (A/B benchmarking; auto-dispatch uses the LDS
+  // selection kernel for small b)
+  auto v = view_of(U);
+  TORCH_CHECK(v.K - 2 * b >= 1 && b > 0, "bad b");
+  auto out = torch::empty({v.d}, U.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const int grid = (int)std::min<long long>(cdiv(v.d, RT_T), kMaxBlocks);
+  radix_trimmed_kernel<<<grid, 256, 0, stream>>>(
+      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b);
+  return out;
+}
+
 torch::Tensor col_median(torch::Tensor U) {
   auto v = view_of(U);
   const long b = (v.K - 1) / 2;  // K-2b = 1 (odd K) or 2 (even K: avg both)
@@ -708,6 +721,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "masked column mean+std (K10)", py::arg("U"), py::arg("mask"),
         py::arg("unbiased") = true, py::arg("count") = -1.0);
   m.def("trimmed_mean", &trimmed_mean, "coordinate-wise trimmed mean (K3)");
+  m.def("trimmed_mean_radix", &trimmed_mean_radix,
+        "trimmed mean via dual radix select (benchmarking entry)");
   m.def("col_median", &col_median, "coordinate-wise median (K2)");
   m.def("row_sq_norms", &row_sq_norms, "per-row squared norms");
   m.def("row_diff_sq_norms", &row_diff_sq_norms,
