@@ -47,6 +47,19 @@ class _BaseAggregator:
     def __call__(self, inputs) -> Tensor:
         raise NotImplementedError
 
+    # ---- coordinate-sharded execution
+    # Row-wise aggregators (GeoMed/Krum/...) override aggregate_shard to run
+    # on a [K, d/ws] coordinate shard: shard-local column math + all-reduce
+    # of the small K-length / K×K row statistics.  Coordinate-wise
+    # aggregators don't need it (their __call__ on the shard IS the shard
+    # of the answer).
+    supports_shard: bool = False
+
+    def aggregate_shard(self, U_shard: Tensor, runtime) -> Tensor:
+        """Return this rank's d/ws slice of the aggregate."""
+        raise NotImplementedError(
+            f"{type(self).__name__} has no shard-aware path")
+
     # ---- checkpointing hooks (stateless default)
     def state_dict(self) -> dict:
         return {}

@@ -42,6 +42,29 @@ class Clippedclustering(_BaseAggregator):
         sel = torch.from_numpy(labels == flag).to(U.device)
         return ops.col_mean(U[sel])
 
+    supports_shard = True
+
+    def aggregate_shard(self, U_shard, runtime):
+        """Distributed form: global row norms from partial sq-norms +
+        all-reduce (the norm history stays replicated — every rank sees the
+        same norms), shard rows clipped by the global threshold, then the
+        shard-aware clustering scheme."""
+        from .clustering import Clustering
+
+        U = U_shard.clone()
+        K = U.shape[0]
+        sqn = ops.row_sq_norms(U)
+        runtime.all_reduce_(sqn)
+        norms = sqn.sqrt()
+        self.l2norm_his.extend(norms.cpu().tolist())
+        threshold = self.tau if self.tau else float(np.median(self.l2norm_his))
+        scale = torch.clamp(threshold / norms.clamp_min(1e-12), max=1.0)
+        U.mul_(scale.unsqueeze(1))
+        G = ops.gram(U)
+        runtime.all_reduce_(G)
+        sel = Clustering._select_larger_cluster(G)
+        return ops.col_mean(U[sel])
+
     def state_dict(self) -> dict:
         return {"l2norm_his": list(self.l2norm_his)}
 

@@ -54,18 +54,34 @@ def complete_linkage_two_clusters(dist: np.ndarray) -> np.ndarray:
 
 
 class Clustering(_BaseAggregator):
-    def __call__(self, inputs):
-        U = self._get_updates(inputs)
-        K = U.shape[0]
-        cos = ops.cos_sim_gram(U)  # [-1, 1]
+    supports_shard = True
+
+    @staticmethod
+    def _select_larger_cluster(G: torch.Tensor) -> torch.Tensor:
+        """2-cluster complete linkage on 1−cos from a full Gram; returns the
+        bool row mask of the larger cluster."""
+        K = G.shape[0]
+        norms = G.diagonal().sqrt().clamp_min(1e-8)
+        cos = (G / norms.unsqueeze(0) / norms.unsqueeze(1)).clamp(-1.0, 1.0)
         sim = cos.cpu().numpy()
         np.fill_diagonal(sim, 1.0)
         sim = np.nan_to_num(sim, nan=-1.0, posinf=1.0, neginf=-1.0)
-        dist = 1.0 - sim
-        labels = complete_linkage_two_clusters(dist)
+        labels = complete_linkage_two_clusters(1.0 - sim)
         flag = 1 if labels.sum() > K // 2 else 0
-        sel = torch.from_numpy(labels == flag).to(U.device)
+        return torch.from_numpy(labels == flag).to(G.device)
+
+    def __call__(self, inputs):
+        U = self._get_updates(inputs)
+        sel = self._select_larger_cluster(ops.gram(U))
         return ops.col_mean(U[sel])
+
+    def aggregate_shard(self, U_shard, runtime):
+        """Partial Gram + all-reduce; clustering (tiny, host-side) is
+        replicated on every rank, the cluster mean stays shard-local."""
+        G = ops.gram(U_shard)
+        runtime.all_reduce_(G)
+        sel = self._select_larger_cluster(G)
+        return ops.col_mean(U_shard[sel])
 
     def __str__(self):
         return "Clustering"

@@ -16,6 +16,8 @@ from .base import _BaseAggregator
 
 
 class Krum(_BaseAggregator):
+    supports_shard = True
+
     def __init__(self, num_clients: int = 20, num_byzantine: int = 5, m: int = 1):
         super().__init__()
         self.n = num_clients
@@ -35,6 +37,28 @@ class Krum(_BaseAggregator):
         scores = ops.krum_scores(D, self.f)
         top_m = scores.argsort()[: self.m]
         return U[top_m].sum(dim=0)
+
+    def aggregate_shard(self, U_shard, runtime):
+        """Distributed Krum (SURVEY.md §5.7/§7 hard-part 3): each rank
+        computes the partial Gram of its coordinate shard (MFMA kernel K4),
+        one all-reduce of the K×K matrix assembles the exact pairwise
+        distances, selection is replicated, and the winning rows are summed
+        shard-locally."""
+        import torch as _t
+
+        from blades_amd import ops
+
+        n = U_shard.shape[0]
+        if 2 * self.f + 2 > n:
+            raise ValueError(f"Too many Byzantine workers: 2*{self.f}+2 > {n}")
+        G = ops.gram(U_shard)
+        runtime.all_reduce_(G)
+        sq = G.diagonal()
+        D = (sq.unsqueeze(0) + sq.unsqueeze(1) - 2 * G).clamp_min_(0)
+        D.fill_diagonal_(0)
+        scores = ops.krum_scores(D, self.f)
+        top_m = scores.argsort()[: self.m]
+        return U_shard[top_m].sum(dim=0)
 
     def __str__(self):
         return f"Krum (m={self.m})"

@@ -193,3 +193,73 @@ def test_coordinate_with_noise_attack_matches_full():
     # trimmed-mean sums reduce in a layout-dependent order (sharded columns
     # are strided views) -> occasional 1-ulp differences
     assert np.allclose(outs["coordinate"], outs["full"], atol=1e-7)
+
+
+AGG_SHARD_CASES = [
+    ("geomed", {}),
+    ("autogm", {"lamb": 1.0}),
+    ("krum", {"num_clients": 8, "num_byzantine": 2}),
+    ("multikrum", {"num_clients": 8, "num_byzantine": 2, "m": 3}),
+    ("clustering", {}),
+    ("clippedclustering", {}),
+]
+
+
+def _run_shard_aggs(gather, seed=11):
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    out = {}
+    for name, kws in AGG_SHARD_CASES:
+        ds = SyntheticFLDataset(num_clients=8, samples_per_client=16,
+                                batch_size=8, shape=(1, 28, 28),
+                                num_classes=10, seed=0)
+        sim = Simulator(ds, num_byzantine=2, attack="ipm",
+                        aggregator=name, aggregator_kws=kws,
+                        log_path=f"/tmp/bl_shagg_{gather}_{name}_"
+                                 f"{os.environ.get('RANK', 's')}",
+                        seed=seed, gather=gather)
+        torch.manual_seed(seed)
+        sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
+        out[name] = sim.server.flat_parameters().numpy().copy()
+    return out
+
+
+def _shard_agg_worker(rank, world_size, port, out_q):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world_size),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    out = _run_shard_aggs("coordinate")
+    out_q.put((rank, out))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_shard_aware_rowwise_aggregators_match_single_rank():
+    """GeoMed/AutoGM/Krum/Multi-Krum/Clustering/ClippedClustering on the
+    coordinate-sharded path (partial norms/Gram + all-reduce) must match
+    the single-rank full computation."""
+    import numpy as np
+
+    single = _run_shard_aggs("full")  # ws=1: full path
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_shard_agg_worker, args=(r, WORLD, 29655, q))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(WORLD):
+        rank, out = q.get(timeout=500)
+        res[rank] = out
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    for name, _ in AGG_SHARD_CASES:
+        assert np.array_equal(res[0][name], res[1][name]), name
+        assert np.allclose(res[0][name], single[name], atol=1e-6), name
