@@ -115,3 +115,14 @@ class SyntheticFLDataset:
             X, y = self._gen_pool(u_id, self.test_samples_per_client, tag=2)
             self._test[u_id] = CustomTensorDataset(X.cpu(), y.cpu())
         return self._test[u_id]
+
+    def get_stacked_test_data(self, ids: Sequence[int], device=None):
+        """[C, n_test, ...] stacked test tensors (batched-eval fast path);
+        identical data to get_all_test_data (same CPU-generated pools)."""
+        dev = torch.device(device) if device is not None else self.device
+        xs, ys = [], []
+        for u in ids:
+            X, y = self._gen_pool(u, self.test_samples_per_client, tag=2)
+            xs.append(X)
+            ys.append(y)
+        return torch.stack(xs).to(dev), torch.stack(ys).to(dev)

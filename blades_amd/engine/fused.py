@@ -317,6 +317,15 @@ class FusedEngine:
         """
         params = dict(self.spec.named_slices(theta))
         self.base.eval()
+        stacked = getattr(dataset, "get_stacked_test_data", None)
+        if stacked is not None and clients:
+            try:
+                out = self._evaluate_stacked(params, stacked, clients,
+                                             round_number, batch_size, metrics)
+                self.base.train()
+                return out
+            except NotImplementedError:
+                pass
         results = []
         for c in clients:
             test_set = dataset.get_all_test_data(c.id())
@@ -338,4 +347,32 @@ class FusedEngine:
             r["Loss"] /= max(r["Length"], 1)
             results.append(r)
         self.base.train()
+        return results
+
+    @torch.no_grad()
+    def _evaluate_stacked(self, params, stacked_fn, clients, round_number,
+                          batch_size, metrics):
+        """Batched eval: all shard clients' test sets in a few big forwards
+        (weights are identical across clients — the per-client loop only
+        cost host time)."""
+        X, Y = stacked_fn([c.id() for c in clients], device=self.device)
+        C, n = X.shape[0], X.shape[1]
+        chunk = max(1, (batch_size * 64) // max(n, 1))
+        outs = []
+        for i in range(0, C, chunk):
+            xb = X[i:i + chunk].flatten(0, 1)
+            o = functional_call(self.base, (params, self.buffers), (xb,))
+            outs.append(o.view(-1, n, o.shape[-1]))
+        O = torch.cat(outs)  # [C, n, num_classes]
+        losses = nn.functional.cross_entropy(
+            O.reshape(C * n, -1), Y.reshape(C * n), reduction="none"
+        ).view(C, n).mean(1)
+        results = []
+        for i, c in enumerate(clients):
+            r = {"_meta": {"type": "client_validation"}, "E": round_number,
+                 "Client": c.id(), "Length": n,
+                 "Loss": float(losses[i].item())}
+            for name, metric in metrics.items():
+                r[name] = metric(O[i], Y[i])
+            results.append(r)
         return results

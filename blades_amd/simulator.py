@@ -503,10 +503,18 @@ class Simulator:
             delta = rt.all_gather_flat(delta_shard)[:d]
             self.server.apply_update(delta)
 
-    # reference-name alias
+    # reference-name aliases.  The reference's "trainer" mode
+    # (simulator.py:249-280, half-built ray.train/DDP pool) is subsumed by
+    # the same fused rank runtime — both names run the same round.
     def train_actor(self, global_round: int, num_rounds: int,
                     clients: List[BladesClient], lr: float) -> None:
         self.train_round(global_round, num_rounds, clients, lr)
+
+    def train_trainer(self, epoch: int, num_rounds: int, clients) -> None:
+        client_list = (list(clients.values()) if isinstance(clients, dict)
+                       else list(clients))
+        lr = getattr(self, "_last_client_lr", 0.1)
+        self.train_round(epoch, num_rounds, client_list, lr)
 
     # --------------------------------------------------------------- eval
     def test_actor(self, global_round: int, batch_size: int):
@@ -631,6 +639,7 @@ class Simulator:
         global_start = time.time()
         ret: List[float] = []
         cur_lr = client_lr
+        self._last_client_lr = client_lr
         for r in range(1, global_rounds + 1):
             round_start = time.time()
             self.train_round(r, local_steps, clients, cur_lr)
@@ -642,6 +651,7 @@ class Simulator:
             if client_lr_scheduler:
                 client_lr_scheduler.step()
                 cur_lr = client_lr_scheduler.get_last_lr()[0]
+                self._last_client_lr = cur_lr
             ret.append(time.time() - round_start)
             self.debug_logger.info(
                 f"E={r}; Client learning rate = {cur_lr}; "

@@ -100,3 +100,36 @@ def test_client_chunking_equivalent():
         eng = FusedEngine(model, spec, torch.device("cpu"), client_chunk=chunk)
         outs.append(eng.run_round(theta.clone(), clients, ds, 1, 0.1))
     assert torch.allclose(outs[0], outs[1], atol=1e-7)
+
+
+def test_stacked_eval_matches_per_client_eval():
+    """Batched eval fast path == per-client DataLoader eval."""
+    import copy
+
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.client import BladesClient
+    from blades_amd.engine import FusedEngine, LoopEngine, ParamSpec
+    from blades_amd.utils import top1_accuracy
+
+    torch.manual_seed(1)
+    model = MLP()
+    spec = ParamSpec.from_module(model)
+    theta = spec.flatten(model)
+    ds = SyntheticFLDataset(num_clients=5, samples_per_client=8, batch_size=4,
+                            shape=(1, 28, 28), num_classes=10, seed=0,
+                            test_samples_per_client=12)
+    clients = [BladesClient(id=i) for i in range(5)]
+    eng = FusedEngine(model, spec, torch.device("cpu"))
+    fast = eng.evaluate(theta, clients, ds, 1, 4, {"top1": top1_accuracy})
+
+    # reference path: loop engine with per-client model
+    for c in clients:
+        c.set_model(model)
+    loop = LoopEngine(device="cpu")
+    slow = loop.evaluate(model, clients, ds, 1, 4, {"top1": top1_accuracy})
+
+    assert len(fast) == len(slow) == 5
+    for f, s in zip(fast, slow):
+        assert f["Length"] == s["Length"]
+        assert abs(f["Loss"] - s["Loss"]) < 1e-5, (f, s)
+        assert abs(f["top1"] - s["top1"]) < 1e-3
