@@ -565,6 +565,28 @@ class Simulator:
         }
         self.json_logger.write(r)
 
+    def log_train(self, progress, batch_idx, epoch, results) -> None:
+        """Weighted training-metric record (reference: simulator.py:337-362
+        — a dead path there due to two bugs; functional here).
+
+        ``results``: list of {"length": n, "loss": l, "metrics": {...}}.
+        """
+        length = sum(res["length"] for res in results)
+        r = {
+            "_meta": {"type": "train"},
+            "Round": epoch,
+            "B": batch_idx,
+            "Length": length,
+            "Loss": sum(res["loss"] * res["length"] for res in results) / length,
+        }
+        for metric_name in self.metrics:
+            r[metric_name] = (sum(res["metrics"][metric_name] * res["length"]
+                                  for res in results) / length)
+        self.debug_logger.info(
+            f"[Round {epoch} B{batch_idx}] Loss: {r['Loss']:.4f} "
+            + " ".join(f"{name}={r[name]:>8.4f}" for name in self.metrics))
+        self.json_logger.write(r)
+
     def log_validate(self, metrics):
         top1 = np.average([m["top1"] for m in metrics],
                           weights=[m["Length"] for m in metrics])

@@ -116,3 +116,23 @@ def test_algorithm_presets(tmp_path):
     ret = sim.run(MLP(), global_rounds=1, validate_interval=0, **fedsgd())
     assert len(ret) == 1
     assert fedavg(local_steps=3)["local_steps"] == 3
+
+
+def test_cli_args_surface():
+    """Reference flag surface parses (scripts/args.py parity) and the
+    log-dir naming scheme matches the reference pattern."""
+    import sys
+    sys.path.insert(0, "scripts")
+    try:
+        from args import parse_arguments
+    finally:
+        sys.path.pop(0)
+    opts = parse_arguments([
+        "--use-cuda", "--seed", "3", "--global_round", "10",
+        "--local_round", "5", "--batch_size", "16", "--attack", "ipm",
+        "--agg", "trimmedmean", "--lr", "0.2", "--num_byzantine", "4",
+        "--num_actors", "20", "--num_gpus", "4", "--dataset", "cifar10",
+    ])
+    assert opts.num_byzantine == 4
+    assert opts.agg_args["trimmedmean"] == {"nb": 4}
+    assert "/b4_ipm_epsilon0.5_trimmedmean_nb4_lr0.2_bz16_seed3" in opts.log_dir

@@ -155,3 +155,19 @@ def test_determinism_same_seed(tmp_path):
         sim.run(MLP(), global_rounds=3, validate_interval=0)
         outs.append(sim.server.flat_parameters())
     assert torch.equal(outs[0], outs[1])
+
+
+def test_log_train_record(tmp_path):
+    import json as _json
+
+    sim = Simulator(make_ds(4), log_path=str(tmp_path), seed=0)
+    sim.run(MLP(), global_rounds=1, validate_interval=0)
+    sim.log_train(progress=10, batch_idx=2, epoch=1, results=[
+        {"length": 4, "loss": 1.0, "metrics": {"top1": 50.0}},
+        {"length": 12, "loss": 2.0, "metrics": {"top1": 25.0}},
+    ])
+    recs = [_json.loads(l) for l in open(tmp_path / "stats")]
+    train = [r for r in recs if r["_meta"]["type"] == "train"]
+    assert len(train) == 1
+    assert abs(train[0]["Loss"] - (4 * 1.0 + 12 * 2.0) / 16) < 1e-9
+    assert abs(train[0]["top1"] - (4 * 50 + 12 * 25) / 16) < 1e-9
