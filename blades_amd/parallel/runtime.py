@@ -103,6 +103,28 @@ class DistributedRuntime:
         blocks = [out[r * kmax: r * kmax + sizes[r]] for r in range(self.world_size)]
         return torch.cat(blocks, dim=0)
 
+    def all_to_all_row_block(self, local: torch.Tensor, counts,
+                             dshard: int) -> torch.Tensor:
+        """Like :meth:`all_to_all_coordinate_shard` but for an explicit
+        per-rank row-count list (streamed client chunks): rank r
+        contributes ``counts[r]`` rows; returns [sum(counts), dshard] in
+        rank order."""
+        if not self.distributed:
+            return local
+        ws = self.world_size
+        kmax = max(counts)
+        if local.shape[0] < kmax:
+            pad = torch.zeros(kmax - local.shape[0], local.shape[1],
+                              device=local.device, dtype=local.dtype)
+            send = torch.cat([local, pad], dim=0)
+        else:
+            send = local
+        send = send.view(kmax, ws, dshard).transpose(0, 1).contiguous()
+        recv = torch.empty_like(send)
+        dist.all_to_all_single(recv, send)
+        blocks = [recv[r, :counts[r]] for r in range(ws)]
+        return torch.cat(blocks, dim=0)
+
     def all_to_all_coordinate_shard(self, local: torch.Tensor,
                                     total_rows: int,
                                     dshard: int) -> torch.Tensor:

@@ -89,6 +89,7 @@ class Simulator:
         device: Optional[str] = None,
         hip_graphs: bool = True,
         gather: str = "auto",
+        stream_clients: Optional[int] = None,
         **kwargs,
     ):
         self.use_actor = mode == "actor"
@@ -168,6 +169,12 @@ class Simulator:
         if gather not in ("auto", "full", "coordinate"):
             raise ValueError(f"gather must be auto|full|coordinate, got {gather}")
         self._gather = gather
+        # streamed coordinate rounds: train + reshard the shard in chunks of
+        # this many clients, so the rank-local client slab never
+        # materializes (1e4-client × WRN scale: slab and coordinate shard
+        # are each 182 GB/rank — together they exceed 288 GB; streamed, the
+        # transient is stream_clients × d floats)
+        self._stream_clients = stream_clients
 
     # ------------------------------------------------------------ builders
     def _init_aggregator(self, aggregator, aggregator_kws) -> None:
@@ -365,6 +372,13 @@ class Simulator:
         else:
             dshard = 0
             d_pad = -(-d // 4) * 4
+
+        if coordinate and self._stream_clients:
+            self._train_round_streamed(global_round, local_steps, lr,
+                                       all_clients, shard, rows, dshard, d,
+                                       d_pad)
+            return
+
         with trace_range("blades/local_train"):
             theta = self.server.flat_parameters(device=self.device,
                                                 out=self._theta)
@@ -466,7 +480,63 @@ class Simulator:
             Ucoord = rt.all_to_all_coordinate_shard(
                 buf_local, total_rows=len(all_clients), dshard=dshard)
             torch.nan_to_num_(Ucoord)
+        self._finish_coordinate(global_round, Ucoord, all_clients, dshard, d)
 
+    def _train_round_streamed(self, global_round: int, local_steps: int, lr,
+                              all_clients, shard, rows, dshard: int, d: int,
+                              d_pad: int) -> None:
+        """Coordinate round in client chunks: train `stream_clients` of the
+        shard, all-to-all that row block, scatter into the coordinate
+        shard, repeat — the rank-local [K/ws, d] slab never materializes
+        (SURVEY.md §7 hard-part 2; docs/roadmap.md item 3)."""
+        from blades_amd.attackers import NoiseClient
+        from blades_amd.ops import philox_normal
+        from blades_amd.utils import client_philox_seed
+
+        rt = self.runtime
+        chunk = self._stream_clients
+        sizes = [len(s) for s in rt.shard_indices(len(all_clients))]
+        offsets = np.cumsum([0] + sizes[:-1])
+        n_chunks = max(-(-s // chunk) for s in sizes)
+        theta = self.server.flat_parameters(device=self.device,
+                                            out=self._theta)
+
+        Ucoord = torch.zeros(len(all_clients), dshard, device=self.device)
+        for i in range(n_chunks):
+            lo = i * chunk
+            my_rows = shard[lo:lo + chunk]
+            counts = [max(0, min(chunk, sizes[r] - lo))
+                      for r in range(rt.world_size)]
+            buf = torch.zeros(len(my_rows), d_pad, device=self.device)
+            with trace_range("blades/local_train"):
+                if my_rows:
+                    self._fused.run_round(theta, my_rows, self.dataset,
+                                          local_steps, lr,
+                                          out=buf[:, :d])
+                for j, c in enumerate(my_rows):
+                    if type(c) is NoiseClient:
+                        seed = client_philox_seed(self._seed, rows[c.id()],
+                                                  global_round, tag=12)
+                        buf[j, :d].copy_(philox_normal(
+                            (d,), c._noise_mean, c._noise_std, seed,
+                            device=self.device))
+                    c.save_update_view(buf[j, :d])
+            with trace_range("blades/reshard"):
+                part = rt.all_to_all_row_block(buf, counts, dshard)
+                # scatter the rank-ordered rows to their global positions
+                gidx = torch.tensor(
+                    [int(offsets[r]) + lo + j
+                     for r in range(rt.world_size)
+                     for j in range(counts[r])], device=self.device)
+                Ucoord.index_copy_(0, gidx, part)
+        torch.nan_to_num_(Ucoord)
+        self._finish_coordinate(global_round, Ucoord, all_clients, dshard, d)
+
+    def _finish_coordinate(self, global_round: int, Ucoord, all_clients,
+                           dshard: int, d: int) -> None:
+        from blades_amd.attackers import AlieClient, IpmClient
+
+        rt = self.runtime
         with trace_range("blades/attack"):
             honest = torch.tensor([not c.is_byzantine() for c in all_clients],
                                   device=self.device)

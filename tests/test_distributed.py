@@ -263,3 +263,54 @@ def test_shard_aware_rowwise_aggregators_match_single_rank():
     for name, _ in AGG_SHARD_CASES:
         assert np.array_equal(res[0][name], res[1][name]), name
         assert np.allclose(res[0][name], single[name], atol=1e-6), name
+
+
+def _stream_worker(rank, world_size, port, out_q, stream):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world_size),
+        "LOCAL_RANK": str(rank), "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port),
+    })
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    ds = SyntheticFLDataset(num_clients=7, samples_per_client=16, batch_size=8,
+                            shape=(1, 28, 28), num_classes=10, seed=0)
+    sim = Simulator(ds, num_byzantine=2, attack="alie",
+                    attack_kws={"num_clients": 7, "num_byzantine": 2},
+                    aggregator="geomed",
+                    log_path=f"/tmp/bl_stream_{stream}_{rank}", seed=6,
+                    gather="coordinate", stream_clients=stream)
+    torch.manual_seed(6)
+    sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
+    out_q.put((rank, sim.server.flat_parameters().numpy().copy()))
+    import torch.distributed as dist
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_streamed_coordinate_matches_unstreamed():
+    """Client-chunked streamed reshard (uneven 7-client shard, chunk 2) ==
+    one-shot coordinate reshard, including a shard-aware aggregator
+    (GeoMed) and noise-free ALIE attackers."""
+    import numpy as np
+
+    outs = {}
+    for i, stream in enumerate((None, 2)):
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_stream_worker,
+                             args=(r, WORLD, 29661 + i, q, stream))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        res = {}
+        for _ in range(WORLD):
+            rank, theta = q.get(timeout=240)
+            res[rank] = theta
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        assert np.array_equal(res[0], res[1])
+        outs[stream] = res[0]
+    assert np.array_equal(outs[None], outs[2])
