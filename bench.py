@@ -65,6 +65,8 @@ def parse_args():
     p.add_argument("--aggregator", type=str, default="trimmedmean")
     p.add_argument("--attack", type=str, default="alie")
     p.add_argument("--client-chunk", type=int, default=None)
+    p.add_argument("--stream-clients", type=int, default=None,
+                   help="streamed coordinate rounds: clients per reshard chunk")
     p.add_argument("--num-classes", type=int, default=10)
     p.add_argument("--samples-per-client", type=int, default=64)
     args, _ = p.parse_known_args()
@@ -85,7 +87,6 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
 
     use_cuda = torch.cuda.is_available()
-    device_kw = {}
     if not use_cuda:
         # CPU smoke mode (scaled down so it finishes in seconds)
         args.clients_per_gpu = min(args.clients_per_gpu, 8)
@@ -132,6 +133,7 @@ def main():
         log_path=os.environ.get("BLADES_BENCH_LOG", "/tmp/blades_bench_logs"),
         seed=1234,
         client_chunk=args.client_chunk,
+        stream_clients=args.stream_clients,
     )
     device = sim.device
     # move the synthetic pools to the device and pre-materialize the shard

@@ -76,7 +76,6 @@ class FusedEngine:
         # buffers (none for vmap-safe BN; kept for generality) are shared
         self.buffers = {k: v.detach().clone() for k, v in self.base.named_buffers()}
         self.client_chunk = client_chunk
-        self._grad_fn = None
         # population path: direct MFMA popconv kernels for supported models.
         # Measured round 1: the popconv round runs ~467 ms vs 229 ms for the
         # vmap+hipGraph path on the headline config (popconv_fwd reaches

@@ -25,7 +25,7 @@ without re-capture.
 """
 from __future__ import annotations
 
-from typing import Callable, List, Optional
+from typing import List, Optional
 
 import torch
 
@@ -38,10 +38,6 @@ from blades_amd.client import BladesClient
 _CAPTURABLE_AGGS = (Mean, Median, Trimmedmean, Krum)
 _CAPTURABLE_BYZ = (AlieClient, IpmClient, LabelflippingClient,
                    SignflippingClient)
-
-
-class Unsupported(Exception):
-    pass
 
 
 class CapturedRound:
