@@ -106,3 +106,68 @@ def test_graph_round_matches_eager():
     te, _ = go(False)
     assert used, "graph capture did not engage"
     assert torch.allclose(tg, te, atol=2e-5), (tg - te).abs().max()
+
+
+def test_scheduler_under_graphs():
+    """LR schedulers must work across graph replays (learning rates live in
+    device scalars; no re-capture)."""
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import resnet18
+
+    def go(graphs):
+        ds = SyntheticFLDataset(num_clients=8, samples_per_client=16,
+                                batch_size=8, shape=(3, 32, 32),
+                                num_classes=10, seed=0, device="cuda:0")
+        sim = Simulator(ds, num_byzantine=2, attack="alie",
+                        attack_kws={"num_clients": 8, "num_byzantine": 2},
+                        aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                        use_cuda=True, log_path=f"/tmp/bl_sched_{graphs}",
+                        seed=2, hip_graphs=graphs)
+        model = resnet18(norm="batch-local")
+        server_opt = None
+        # client scheduler halves lr at rounds 3 and 5
+        class FakeSched:
+            def __init__(self):
+                self.lr = 0.02
+                self.n = 0
+            def step(self):
+                self.n += 1
+                if self.n in (3, 5):
+                    self.lr /= 2
+            def get_last_lr(self):
+                return [self.lr]
+        ret = sim.run(model, global_rounds=6, local_steps=1, client_lr=0.02,
+                      server_lr=1.0, validate_interval=0,
+                      client_lr_scheduler=FakeSched())
+        used = sim._graph_round is not None and sim._graph_round.graph is not None
+        return sim.server.flat_parameters().cpu(), used
+
+    tg, used = go(True)
+    te, _ = go(False)
+    assert used
+    assert torch.allclose(tg, te, atol=5e-4), (tg - te).abs().max()
+
+
+def test_mixed_population_gpu():
+    """Custom client subclass (loop engine) + fused rest on GPU."""
+    from blades_amd import BladesClient, Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import MLP
+
+    seen = []
+
+    class MyClient(BladesClient):
+        def local_training(self, data_batches):
+            seen.append(self.id())
+            super().local_training(data_batches)
+
+    ds = SyntheticFLDataset(num_clients=6, samples_per_client=16, batch_size=8,
+                            shape=(1, 28, 28), num_classes=10, seed=0,
+                            device="cuda:0")
+    sim = Simulator(ds, use_cuda=True, log_path="/tmp/bl_mixed_gpu", seed=0)
+    c = MyClient(id=2, device="cuda:0")
+    sim._clients[2] = c
+    sim.run(MLP(), global_rounds=2, validate_interval=0)
+    assert seen == [2, 2]
+    assert torch.isfinite(sim.server.flat_parameters()).all()
