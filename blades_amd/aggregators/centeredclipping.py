@@ -29,10 +29,46 @@ class Centeredclipping(_BaseAggregator):
         if self.momentum is None:
             self.momentum = torch.zeros(U.shape[1], device=U.device, dtype=U.dtype)
         else:
+            if self.momentum.numel() > U.shape[1]:
+                # momentum from a coordinate-sharded run carries zero pad
+                # columns at the tail; trim to the true d
+                self.momentum = self.momentum[:U.shape[1]]
             self.momentum = self.momentum.to(device=U.device, dtype=U.dtype)
         for _ in range(self.n_iter):
             self.momentum = ops.centered_clip_iter(U, self.momentum, self.tau)
         return self.momentum.clone().detach()
+
+    supports_shard = True
+
+    def aggregate_shard(self, U_shard: torch.Tensor, runtime) -> torch.Tensor:
+        """Distributed centered clipping: per-iteration row norms from
+        partial sq-norms + all-reduce; the momentum vector is kept FULL on
+        every rank (d floats — tiny next to the slab) with only this rank's
+        slice updated per iteration and re-assembled by one all-gather at
+        the end, so checkpointing stays rank-independent."""
+        from blades_amd import ops
+
+        K, dshard = U_shard.shape
+        full_len = dshard * runtime.world_size
+        if self.momentum is None or self.momentum.numel() != full_len:
+            # (re)initialize in the padded full length; slices align with
+            # the coordinate shards
+            self.momentum = torch.zeros(full_len, device=U_shard.device,
+                                        dtype=U_shard.dtype)
+        else:
+            self.momentum = self.momentum.to(device=U_shard.device,
+                                             dtype=U_shard.dtype)
+        lo = runtime.rank * dshard
+        v_shard = self.momentum[lo:lo + dshard].clone()
+        for _ in range(self.n_iter):
+            part = ops.row_diff_norms(U_shard, v_shard) ** 2
+            runtime.all_reduce_(part)
+            norms = part.sqrt().clamp_min(1e-12)
+            scale = torch.clamp(self.tau / norms, max=1.0) / K
+            v_shard = v_shard * (1.0 - scale.sum()) \
+                + ops.weighted_col_sum(U_shard, scale)
+        self.momentum = runtime.all_gather_flat(v_shard)
+        return v_shard
 
     def state_dict(self) -> dict:
         return {"momentum": None if self.momentum is None else self.momentum.cpu()}

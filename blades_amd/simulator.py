@@ -373,7 +373,10 @@ class Simulator:
             dshard = 0
             d_pad = -(-d // 4) * 4
 
-        if coordinate and self._stream_clients:
+        if (coordinate and self._stream_clients
+                and self._engine_choice != "loop"):
+            # streamed rounds drive the fused engine directly (the loop
+            # engine would defeat the memory point of streaming)
             self._train_round_streamed(global_round, local_steps, lr,
                                        all_clients, shard, rows, dshard, d,
                                        d_pad)
@@ -567,7 +570,14 @@ class Simulator:
             if getattr(self.aggregator, "coordinate_shardable", False):
                 delta_shard = self.aggregator(Ucoord)
             else:  # row-wise aggregator with a shard-aware form
-                delta_shard = self.aggregator.aggregate_shard(Ucoord, rt)
+                import inspect
+
+                sig = inspect.signature(self.aggregator.aggregate_shard)
+                if "clients" in sig.parameters:
+                    delta_shard = self.aggregator.aggregate_shard(
+                        Ucoord, rt, clients=all_clients)
+                else:
+                    delta_shard = self.aggregator.aggregate_shard(Ucoord, rt)
 
         with trace_range("blades/apply"):
             delta = rt.all_gather_flat(delta_shard)[:d]

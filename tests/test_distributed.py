@@ -202,6 +202,8 @@ AGG_SHARD_CASES = [
     ("multikrum", {"num_clients": 8, "num_byzantine": 2, "m": 3}),
     ("clustering", {}),
     ("clippedclustering", {}),
+    ("centeredclipping", {"tau": 5.0, "n_iter": 3}),
+    ("fltrust", {}),
 ]
 
 
@@ -219,6 +221,8 @@ def _run_shard_aggs(gather, seed=11):
                         log_path=f"/tmp/bl_shagg_{gather}_{name}_"
                                  f"{os.environ.get('RANK', 's')}",
                         seed=seed, gather=gather)
+        if name == "fltrust":
+            sim.set_trusted_clients([7])
         torch.manual_seed(seed)
         sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
         out[name] = sim.server.flat_parameters().numpy().copy()
