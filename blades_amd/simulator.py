@@ -373,6 +373,25 @@ class Simulator:
             dshard = 0
             d_pad = -(-d // 4) * 4
 
+        if (coordinate and self._stream_clients is None
+                and self._engine_choice != "loop"
+                and self.device.type == "cuda"):
+            # memory-aware default: stream when slab + coordinate shard
+            # would not both fit comfortably (the reshard transiently holds
+            # the client slab [K/ws, d] AND the coordinate shard [K, d/ws]
+            # — each K·d/ws·4 bytes)
+            try:
+                free_b, _ = torch.cuda.mem_get_info(self.device)
+                both = 2 * len(shard) * d_pad * 4
+                if both > 0.5 * free_b:
+                    auto = max(1, int((0.2 * free_b) / (d_pad * 4)))
+                    self._stream_clients = auto
+                    self.debug_logger.info(
+                        f"auto stream_clients={auto} (slab+shard {both/1e9:.1f} "
+                        f"GB vs {free_b/1e9:.1f} GB free)")
+            except Exception:
+                pass
+
         if (coordinate and self._stream_clients
                 and self._engine_choice != "loop"):
             # streamed rounds drive the fused engine directly (the loop

@@ -171,3 +171,28 @@ def test_mixed_population_gpu():
     sim.run(MLP(), global_rounds=2, validate_interval=0)
     assert seen == [2, 2]
     assert torch.isfinite(sim.server.flat_parameters()).all()
+
+
+def test_graph_round_fedavg_matches_eager():
+    """hipGraph capture with local_steps > 1 (FedAvg slab path in-graph)."""
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import MLP
+
+    def go(graphs):
+        ds = SyntheticFLDataset(num_clients=8, samples_per_client=32,
+                                batch_size=8, shape=(1, 28, 28),
+                                num_classes=10, seed=0, device="cuda:0")
+        sim = Simulator(ds, num_byzantine=2, attack="ipm",
+                        aggregator="median", use_cuda=True,
+                        log_path=f"/tmp/bl_gfa_{graphs}", seed=4,
+                        hip_graphs=graphs)
+        sim.run(MLP(), global_rounds=4, local_steps=3, client_lr=0.05,
+                server_lr=1.0, validate_interval=0)
+        used = sim._graph_round is not None and sim._graph_round.graph is not None
+        return sim.server.flat_parameters().cpu(), used
+
+    tg, used = go(True)
+    te, _ = go(False)
+    assert used
+    assert torch.allclose(tg, te, atol=1e-5)
