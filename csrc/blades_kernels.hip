@@ -234,10 +234,34 @@ __global__ void trimmed_select_kernel(const float* __restrict__ U,
     // unroll issues independent loads together — the 4 B/lane column
     // stream is latency-bound otherwise (measured 8.7 ms vs the 0.7 ms
     // HBM floor at K=100, d=11.2M without it).
-    for (; k + 8 <= K; k += 8) {
+    if (k + 8 <= K) {
+      // software pipeline: group g+1's loads issue BEFORE group g's
+      // branchy heap maintenance (hipcc otherwise fences loads behind the
+      // data-dependent LDS branches — guide §5.4 rule 4c)
       float v8[8];
       #pragma unroll
       for (int u = 0; u < 8; ++u) v8[u] = U[(k + u) * ld + j];
+      k += 8;
+      for (; k + 8 <= K; k += 8) {
+        float n8[8];
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) n8[u] = U[(k + u) * ld + j];
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const float v = v8[u];
+          sum += v;
+          if (v < lo_root) {
+            sift_down_max(lo, BS, b, v);
+            lo_root = lo[0];
+          }
+          if (v > hi_root) {
+            sift_down_min(hi, BS, b, v);
+            hi_root = hi[0];
+          }
+        }
+        #pragma unroll
+        for (int u = 0; u < 8; ++u) v8[u] = n8[u];
+      }
       #pragma unroll
       for (int u = 0; u < 8; ++u) {
         const float v = v8[u];

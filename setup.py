@@ -16,7 +16,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 setup(
     name="blades_amd",
     version="0.1.0",
-    packages=["blades_amd"],
+    packages=["blades_amd", "blades"],
     ext_modules=[
         CUDAExtension(
             name="blades_amd._hip_ops",

@@ -136,3 +136,29 @@ def test_cli_args_surface():
     assert opts.num_byzantine == 4
     assert opts.agg_args["trimmedmean"] == {"nb": 4}
     assert "/b4_ipm_epsilon0.5_trimmedmean_nb4_lr0.2_bz16_seed3" in opts.log_dir
+
+
+def test_blades_alias_package():
+    """Reference import paths work verbatim and share class identity."""
+    import blades
+    from blades.simulator import Simulator as S2
+    from blades.client import BladesClient as C2
+    from blades.datasets import SyntheticFLDataset as D2
+    import blades.aggregators.median as med2
+    import blades.attackers.alieclient as alie2
+
+    from blades_amd import Simulator as S1
+    from blades_amd.client import BladesClient as C1
+    from blades_amd.aggregators.median import Median as M1
+    from blades_amd.attackers.alieclient import AlieClient as A1
+
+    assert S2 is S1 and C2 is C1
+    assert med2.Median is M1
+    assert alie2.AlieClient is A1
+    assert blades.Simulator is S1
+
+    # reference-style importlib resolution against the alias
+    import importlib
+    agg_mod = importlib.import_module("blades.aggregators.trimmedmean")
+    assert agg_mod.Trimmedmean is importlib.import_module(
+        "blades_amd.aggregators.trimmedmean").Trimmedmean
