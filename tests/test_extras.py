@@ -162,3 +162,10 @@ def test_blades_alias_package():
     agg_mod = importlib.import_module("blades.aggregators.trimmedmean")
     assert agg_mod.Trimmedmean is importlib.import_module(
         "blades_amd.aggregators.trimmedmean").Trimmedmean
+
+    # the reference's model import paths (scripts/cifar10.py style)
+    from blades.models.cifar10 import CCTNet
+    from blades.models.mnist import MLP as MnistMLP
+    import torch as _t
+    assert CCTNet()( _t.randn(1, 3, 32, 32)).shape == (1, 10)
+    assert MnistMLP()(_t.randn(1, 1, 28, 28)).shape == (1, 10)
