@@ -250,12 +250,17 @@ __global__ void trimmed_select_kernel(const float* __restrict__ U,
         for (int u = 0; u < 8; ++u) {
           const float v = v8[u];
           sum += v;
-          if (v < lo_root) {
-            sift_down_max(lo, BS, b, v);
+          // non-divergent maintenance: when ANY lane replaces, every lane
+          // sifts — non-replacing lanes sift the current root back in (an
+          // identity: the root is the extremum, so it settles at slot 0
+          // immediately).  The divergent form serialized a sparse sift on
+          // nearly every row (7.5 ms vs 0.7 ms floor at K=100, b=20).
+          if (__any(v < lo_root)) {
+            sift_down_max(lo, BS, b, (v < lo_root) ? v : lo_root);
             lo_root = lo[0];
           }
-          if (v > hi_root) {
-            sift_down_min(hi, BS, b, v);
+          if (__any(v > hi_root)) {
+            sift_down_min(hi, BS, b, (v > hi_root) ? v : hi_root);
             hi_root = hi[0];
           }
         }
