@@ -456,6 +456,10 @@ class Simulator:
                     row.copy_(saved.to(self.device))
                     c.save_update_view(row)
             self._attack_ctx = None
+            # K18 again: the reference sanitizes at READ time, i.e. AFTER
+            # omniscient attacks (client.py:198 via _get_updates) — crafted
+            # NaN/Inf rows must not reach the aggregator
+            torch.nan_to_num_(U)
 
         with trace_range("blades/aggregate"):
             agg = self.server.aggregator

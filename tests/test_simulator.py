@@ -171,3 +171,22 @@ def test_log_train_record(tmp_path):
     assert len(train) == 1
     assert abs(train[0]["Loss"] - (4 * 1.0 + 12 * 2.0) / 16) < 1e-9
     assert abs(train[0]["top1"] - (4 * 50 + 12 * 25) / 16) < 1e-9
+
+
+def test_nan_update_is_sanitized(tmp_path):
+    """K18: a client producing NaN/Inf updates must not poison the
+    aggregate (reference semantics: nan_to_num on update read,
+    client.py:198)."""
+    class NaNClient(ByzantineClient):
+        def omniscient_callback(self, simulator):
+            cur = self.get_update()
+            bad = torch.full_like(cur, float("nan"))
+            bad[::2] = float("inf")
+            self._state["saved_update"] = bad
+
+    sim = Simulator(make_ds(6), aggregator="mean", log_path=str(tmp_path),
+                    seed=0)
+    sim.register_attackers([NaNClient()])
+    sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
+    theta = sim.server.flat_parameters()
+    assert torch.isfinite(theta).all()
