@@ -170,6 +170,10 @@ def main():
     rounds_per_sec = args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    from blades_amd.utils.tracing import phase_seconds
+    phases = {k.replace("blades/", ""): round(v, 3)
+              for k, v in sorted(phase_seconds.items())}
+
     if rank == 0:
         print(json.dumps({
             "metric": f"global rounds/sec ({args.model} FedSGD, "
@@ -197,6 +201,7 @@ def main():
                 "seq_len": None,
                 "parallelism": f"client-sharded dp{n_gpus}",
             },
+            "phase_seconds_total": phases,
         }))
 
 
