@@ -179,3 +179,28 @@ def test_pop_engine_gpu_matches_vmap():
     b = run(True)
     assert torch.allclose(a, b, atol=1e-3)
     assert (a - b).abs().mean() < 5e-6
+
+
+@pytest.mark.gpu
+def test_popconv_dw_kernel_matches_bmm():
+    """The split-K dW kernel (kept alongside the shifted-bmm default) must
+    agree with it."""
+    import blades_amd._hip_popconv as ext
+
+    torch.manual_seed(7)
+    C, ci, co, B, H = 3, 32, 64, 4, 16
+    Hp = Wp = H + 2
+    Np = B * Hp * Wp
+    X = torch.randn(C, ci, Np, device="cuda")
+    dY = torch.randn(C, co, Np, device="cuda")
+    dW_k = ext.popconv_dw(dY, X, B, Hp, Wp, False)
+
+    dW_ref = X.new_zeros(C, co, ci, 3, 3)
+    for t in range(9):
+        dy, dx = t // 3 - 1, t % 3 - 1
+        delta = dy * Wp + dx
+        a, b = max(0, -delta), Np - max(0, delta)
+        dW_ref[:, :, :, t // 3, t % 3] = torch.bmm(
+            dY[:, :, a:b], X[:, :, a + delta:b + delta].transpose(1, 2))
+    assert torch.allclose(dW_k, dW_ref, rtol=1e-3, atol=5e-2), \
+        (dW_k - dW_ref).abs().max()
