@@ -228,10 +228,12 @@ torch::Tensor popconv_fwd(torch::Tensor X, torch::Tensor W, long B,
   const long long co = W.size(1);
   TORCH_CHECK(W.size(2) == ci, "ci mismatch");
   TORCH_CHECK(Np == B * Hp * Wp, "Np mismatch");
-  // shared weights arrive as a stride-0 expand over C
+  // W's client dim may be: stride-0 (shared θ expand), contiguous, or the
+  // flat-slab row stride d (ParamSpec batched views) — the kernel only
+  // needs the per-client inner layout contiguous
   long long strideWc = W.stride(0);
-  TORCH_CHECK(strideWc == 0 || strideWc == co * ci * 9,
-              "W must be contiguous per client (or stride-0 broadcast)");
+  TORCH_CHECK(strideWc == 0 || strideWc >= co * ci * 9,
+              "W client stride must be 0 (broadcast) or >= co*ci*9");
   TORCH_CHECK(W.stride(1) == ci * 9 && W.stride(4) == 1,
               "W inner layout must be contiguous");
 

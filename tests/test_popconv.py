@@ -204,3 +204,41 @@ def test_popconv_dw_kernel_matches_bmm():
             dY[:, :, a:b], X[:, :, a + delta:b + delta].transpose(1, 2))
     assert torch.allclose(dW_k, dW_ref, rtol=1e-3, atol=5e-2), \
         (dW_k - dW_ref).abs().max()
+
+
+@pytest.mark.gpu
+def test_pop_engine_gpu_fedavg_matches_vmap():
+    """Population path with divergent per-client weights (FedAvg slab views
+    feed popconv with the flat-slab client stride)."""
+    import os as _os
+
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+
+    def run(pop):
+        env = dict(BLADES_AMD_POPCONV="1") if pop else {}
+        old = {k: _os.environ.get(k) for k in env}
+        _os.environ.update(env)
+        try:
+            torch.manual_seed(8)
+            ds = SyntheticFLDataset(num_clients=6, samples_per_client=16,
+                                    batch_size=8, shape=(3, 32, 32),
+                                    num_classes=10, seed=0, device="cuda:0")
+            sim = Simulator(ds, aggregator="mean", use_cuda=True,
+                            log_path=f"/tmp/bl_popfa_{pop}", seed=9,
+                            hip_graphs=False)
+            sim.run(resnet18(norm="batch-local"), global_rounds=2,
+                    local_steps=2, client_lr=0.005, server_lr=1.0,
+                    validate_interval=0)
+            return sim.server.flat_parameters().cpu()
+        finally:
+            for k, v in old.items():
+                if v is None:
+                    _os.environ.pop(k, None)
+                else:
+                    _os.environ[k] = v
+
+    a = run(True)
+    b = run(False)
+    assert torch.allclose(a, b, atol=1e-3)
+    assert (a - b).abs().mean() < 5e-6
