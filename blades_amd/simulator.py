@@ -407,7 +407,14 @@ class Simulator:
             fusable, custom = split_fusable(shard)
             if self._engine_choice == "loop":
                 fusable, custom = [], shard
-            buf_local = torch.zeros(len(shard), d_pad, device=self.device)
+            # buffer reused across eager rounds (pads were zeroed once and
+            # are never written; the engine overwrites [:, :d] fully)
+            cache_key = (len(shard), d_pad)
+            if getattr(self, "_buf_cache_key", None) != cache_key:
+                self._buf_cache = torch.zeros(len(shard), d_pad,
+                                              device=self.device)
+                self._buf_cache_key = cache_key
+            buf_local = self._buf_cache
             U_local = buf_local[:, :d]
             local_pos = {c.id(): i for i, c in enumerate(shard)}
             if len(fusable) == len(shard):
