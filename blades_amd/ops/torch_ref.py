@@ -34,15 +34,24 @@ def col_median(U: Tensor) -> Tensor:
 
 def trimmed_mean(U: Tensor, b: int) -> Tensor:
     """K3 — per coordinate: drop the b largest and b smallest, mean the rest
-    (reference: aggregators/trimmedmean.py:38-41)."""
+    (reference: aggregators/trimmedmean.py:38-41).
+
+    Accumulated in fp64: the reference's fp32 ``sum − topk + neg_topk`` form
+    cancels catastrophically when Byzantine rows contain huge values (a 1e8
+    outlier swallows every honest value's bits before being subtracted —
+    caught by tests/test_properties.py::test_single_outlier_bounded_influence;
+    the HIP kernel sums in fp64 for the same reason).
+    """
     K = U.shape[0]
     if K - 2 * b <= 0:
         raise ValueError(f"trimmed_mean needs K > 2b (K={K}, b={b})")
     if b == 0:
         return U.mean(dim=0)
-    largest, _ = torch.topk(U, b, 0)
-    neg_smallest, _ = torch.topk(-U, b, 0)
-    return (U.sum(0) - largest.sum(0) + neg_smallest.sum(0)) / (K - 2 * b)
+    Ud = U.double()
+    largest, _ = torch.topk(Ud, b, 0)
+    neg_smallest, _ = torch.topk(-Ud, b, 0)
+    out = (Ud.sum(0) - largest.sum(0) + neg_smallest.sum(0)) / (K - 2 * b)
+    return out.to(U.dtype)
 
 
 def col_var(U: Tensor, unbiased: bool = False) -> Tensor:

@@ -191,3 +191,18 @@ def test_col_median_large_K(ext):
         U = randU(K, 30000, seed=K)
         ref = torch_ref.col_median(U.double()).float()
         assert torch.allclose(ext.col_median(U), ref, atol=1e-6), K
+
+
+def test_trimmed_mean_huge_outlier(ext):
+    """Byzantine-magnitude outliers (1e8) must not corrupt the trimmed mean
+    (fp32 cancellation regression — the reference's formula fails this)."""
+    from blades_amd.ops import torch_ref
+    U = randU(20, 8192, seed=42)
+    U[0] = 1e8
+    U[1] = -1e8
+    out = ext.trimmed_mean(U, 5)
+    ref = torch_ref.trimmed_mean(U.double(), 5).float()
+    assert torch.allclose(out, ref, atol=1e-5, rtol=1e-5)
+    lo = U[2:].min(0).values
+    hi = U[2:].max(0).values
+    assert (out >= lo - 1e-4).all() and (out <= hi + 1e-4).all()
