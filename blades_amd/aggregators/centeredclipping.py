@@ -50,14 +50,25 @@ class Centeredclipping(_BaseAggregator):
 
         K, dshard = U_shard.shape
         full_len = dshard * runtime.world_size
-        if self.momentum is None or self.momentum.numel() != full_len:
-            # (re)initialize in the padded full length; slices align with
-            # the coordinate shards
+        if self.momentum is None:
             self.momentum = torch.zeros(full_len, device=U_shard.device,
                                         dtype=U_shard.dtype)
         else:
             self.momentum = self.momentum.to(device=U_shard.device,
                                              dtype=U_shard.dtype)
+            if self.momentum.numel() != full_len:
+                # momentum carried over from a different world size / gather
+                # layout (e.g. full-gather ↔ coordinate-shard, or resume at
+                # another ws): the first true-d coordinates are the state,
+                # everything past them is zero pad — re-layout, don't drop.
+                m = self.momentum
+                if m.numel() < full_len:
+                    m = torch.cat([m, torch.zeros(full_len - m.numel(),
+                                                  device=m.device,
+                                                  dtype=m.dtype)])
+                else:
+                    m = m[:full_len].clone()
+                self.momentum = m
         lo = runtime.rank * dshard
         v_shard = self.momentum[lo:lo + dshard].clone()
         for _ in range(self.n_iter):

@@ -26,11 +26,9 @@ class Krum(_BaseAggregator):
 
     def __call__(self, inputs):
         U = self._get_updates(inputs)
+        # scores always use the actual population size, not the configured
+        # one (the reference would silently mis-score on a mismatch)
         n = U.shape[0]
-        if n != self.n:
-            # keep running when the caller passes a different population than
-            # configured (the reference would silently mis-score instead)
-            n = U.shape[0]
         if 2 * self.f + 2 > n:
             raise ValueError(f"Too many Byzantine workers: 2*{self.f}+2 > {n}")
         D = ops.pairwise_sq_dists(U)

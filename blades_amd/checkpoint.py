@@ -68,7 +68,12 @@ def load_checkpoint(path: str, model: torch.nn.Module,
                     aggregator=None,
                     restore_rng: bool = True,
                     device: Optional[torch.device] = None) -> Dict[str, Any]:
-    ckpt = torch.load(path, map_location="cpu", weights_only=False)
+    # weights_only load: checkpoints contain only tensors, primitives and
+    # the numpy/python RNG-state tuples — no arbitrary pickle execution.
+    with torch.serialization.safe_globals(
+            [np.ndarray, np.dtype, np.dtypes.UInt32DType,
+             np._core.multiarray._reconstruct]):
+        ckpt = torch.load(path, map_location="cpu", weights_only=True)
     if ckpt.get("format_version") != FORMAT_VERSION:
         raise ValueError(f"unsupported checkpoint version {ckpt.get('format_version')}")
     spec = ParamSpec.from_module(model)

@@ -356,7 +356,19 @@ class FusedEngine:
         cost host time)."""
         X, Y = stacked_fn([c.id() for c in clients], device=self.device)
         C, n = X.shape[0], X.shape[1]
-        chunk = max(1, (batch_size * 64) // max(n, 1))
+        # BN in batch-stats mode (make_vmap_safe strips running stats) would
+        # compute eval statistics across whichever clients share a chunk —
+        # logits would depend on chunk size and neighbours.  Evaluate one
+        # client per forward in that case so stats stay within a client's
+        # own shard (deterministic, chunk-independent).
+        has_batch_stats_bn = any(
+            isinstance(m, nn.modules.batchnorm._BatchNorm)
+            and m.running_mean is None
+            for m in self.base.modules())
+        if has_batch_stats_bn:
+            chunk = 1
+        else:
+            chunk = max(1, (batch_size * 64) // max(n, 1))
         outs = []
         for i in range(0, C, chunk):
             xb = X[i:i + chunk].flatten(0, 1)
