@@ -37,7 +37,12 @@ class DistributedRuntime:
         self.local_rank = int(os.environ.get("LOCAL_RANK", str(self.rank)))
 
         if self.world_size > 1 and not dist.is_initialized():
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            # BLADES_AMD_BACKEND=gloo forces gloo with GPU compute — the
+            # hardware rehearsal mode for multi-process rounds on a 1-GPU
+            # lease (collectives staged through host memory)
+            backend = os.environ.get(
+                "BLADES_AMD_BACKEND",
+                "nccl" if torch.cuda.is_available() else "gloo")
             dist.init_process_group(
                 backend=backend,
                 timeout=datetime.timedelta(seconds=timeout_s),
@@ -56,6 +61,15 @@ class DistributedRuntime:
     def distributed(self) -> bool:
         return self.world_size > 1
 
+    def _stage_cpu(self, t: torch.Tensor) -> bool:
+        """True when a collective on ``t`` must be staged through host
+        memory: the gloo backend cannot move CUDA tensors.  This makes
+        multi-process rounds with GPU compute runnable under gloo — the
+        hardware rehearsal mode for the RCCL path on a 1-GPU lease (RCCL
+        refuses two ranks on one device: 'Duplicate GPU detected')."""
+        return (self.distributed and t.is_cuda
+                and dist.get_backend() == "gloo")
+
     def is_main(self) -> bool:
         return self.rank == 0
 
@@ -73,10 +87,32 @@ class DistributedRuntime:
         idx = self.shard_indices(len(items))[self.rank]
         return [items[i] for i in idx]
 
+    # --------------------------------------------- staged primitive wrappers
+    def _all_gather_into(self, out: torch.Tensor, send: torch.Tensor) -> None:
+        if self._stage_cpu(send):
+            host_out = torch.empty(out.shape, dtype=out.dtype)
+            dist.all_gather_into_tensor(host_out, send.cpu())
+            out.copy_(host_out)
+        else:
+            dist.all_gather_into_tensor(out, send)
+
+    def _all_to_all(self, recv: torch.Tensor, send: torch.Tensor) -> None:
+        if self._stage_cpu(send):
+            host_recv = torch.empty(recv.shape, dtype=recv.dtype)
+            dist.all_to_all_single(host_recv, send.cpu())
+            recv.copy_(host_recv)
+        else:
+            dist.all_to_all_single(recv, send)
+
     # ---------------------------------------------------------- collectives
     def broadcast_flat(self, vec: torch.Tensor, src: int = 0) -> torch.Tensor:
         if self.distributed:
-            dist.broadcast(vec, src=src)
+            if self._stage_cpu(vec):
+                host = vec.cpu()
+                dist.broadcast(host, src=src)
+                vec.copy_(host)
+            else:
+                dist.broadcast(vec, src=src)
         return vec
 
     def all_gather_rows(self, local: torch.Tensor, total_rows: int) -> torch.Tensor:
@@ -99,7 +135,7 @@ class DistributedRuntime:
             send = local.contiguous()
         out = torch.empty(self.world_size * kmax, d, device=local.device,
                           dtype=local.dtype)
-        dist.all_gather_into_tensor(out, send)
+        self._all_gather_into(out, send)
         blocks = [out[r * kmax: r * kmax + sizes[r]] for r in range(self.world_size)]
         return torch.cat(blocks, dim=0)
 
@@ -121,7 +157,7 @@ class DistributedRuntime:
             send = local
         send = send.view(kmax, ws, dshard).transpose(0, 1).contiguous()
         recv = torch.empty_like(send)
-        dist.all_to_all_single(recv, send)
+        self._all_to_all(recv, send)
         blocks = [recv[r, :counts[r]] for r in range(ws)]
         return torch.cat(blocks, dim=0)
 
@@ -151,7 +187,7 @@ class DistributedRuntime:
         # [kmax, ws, dshard] -> [ws, kmax, dshard] so split s goes to rank s
         send = send.view(kmax, ws, dshard).transpose(0, 1).contiguous()
         recv = torch.empty_like(send)
-        dist.all_to_all_single(recv, send)
+        self._all_to_all(recv, send)
         # recv[r] = rank r's rows (padded), MY coordinate slice
         blocks = [recv[r, :sizes[r]] for r in range(ws)]
         return torch.cat(blocks, dim=0)  # [K, dshard], global client order
@@ -162,12 +198,18 @@ class DistributedRuntime:
             return shard
         out = torch.empty(self.world_size * shard.numel(),
                           device=shard.device, dtype=shard.dtype)
-        dist.all_gather_into_tensor(out, shard.contiguous())
+        self._all_gather_into(out, shard.contiguous())
         return out
 
     def all_reduce_(self, t: torch.Tensor, op: str = "sum") -> torch.Tensor:
         if self.distributed:
-            dist.all_reduce(t, op=dist.ReduceOp.SUM if op == "sum" else dist.ReduceOp.MAX)
+            red = dist.ReduceOp.SUM if op == "sum" else dist.ReduceOp.MAX
+            if self._stage_cpu(t):
+                host = t.cpu()
+                dist.all_reduce(host, op=red)
+                t.copy_(host)
+            else:
+                dist.all_reduce(t, op=red)
         return t
 
     def all_gather_object(self, obj) -> List:

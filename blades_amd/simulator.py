@@ -311,8 +311,21 @@ class Simulator:
             self._use_graphs = False
             return False
         server_lr = self.server_opt.param_groups[0]["lr"]
-        with trace_range("blades/graph_round"):
-            self._graph_round.run(lr, server_lr)
+        from blades_amd.engine.graphs import CaptureFailed
+
+        try:
+            with trace_range("blades/graph_round"):
+                self._graph_round.run(lr, server_lr)
+        except CaptureFailed as e:
+            # capture failed (e.g. a backend without stream-capture
+            # support): symmetric across ranks for backend capability
+            # errors — fall back to the eager round
+            self.debug_logger.warning(
+                f"hipGraph capture failed ({e}); falling back to eager")
+            self._graph_round = None
+            self._use_graphs = False
+            torch.cuda.synchronize(self.device)
+            return False
         self._model_stale = True
         return True
 
