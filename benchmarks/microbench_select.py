@@ -9,6 +9,7 @@ def t(f, n=20, w=3):
     torch.cuda.synchronize(); return (time.perf_counter()-t0)/n*1000
 
 for K, b, d in [(100, 20, 11173962), (100, 49, 11173962), (1000, 499, 11173962),
+                (1000, 20, 11173962), (10000, 20, 1400000),
                 (1000, 300, 1400000), (10000, 4999, 1400000)]:
     # padded row stride, as the runtime allocates (float4 kernel paths)
     d_pad = (d + 3) // 4 * 4

@@ -25,7 +25,6 @@ from .torch_ref import (  # ops that are torch on both devices
     col_var,
     clip_to_norm,
     row_clip_to_norm_,
-    krum_scores,
     philox_normal,
     flat_sgd_step_,
     sanitize_,
@@ -94,6 +93,43 @@ def col_median(U: Tensor) -> Tensor:
     if ext is not None:
         return ext.col_median(_prep(U))
     return _ref.col_median(U)
+
+
+def col_trimmed_sum(U: Tensor, b_lo: int, b_hi: int) -> Tensor:
+    """Per-column sum with the b_lo smallest and b_hi largest dropped."""
+    ext = _route(U)
+    if ext is not None:
+        return ext.col_trimmed_sum(_prep(U), b_lo, b_hi)
+    K = U.shape[0]
+    s = U.double().sum(dim=0)
+    if b_lo:
+        s -= torch.topk(U.double(), b_lo, dim=0, largest=False).values.sum(0)
+    if b_hi:
+        s -= torch.topk(U.double(), b_hi, dim=0, largest=True).values.sum(0)
+    return s.to(U.dtype)
+
+
+def krum_scores(sqdists: Tensor, f: int) -> Tensor:
+    """K5 — per-row sum of the n−f−2 smallest squared distances to OTHER
+    rows (reference: aggregators/krum.py:9-25).
+
+    GPU path: sqdists is symmetric with zero diagonal, so row i's score is
+    its COLUMN sum of the smallest n−f−1 entries including the diagonal 0
+    = col_sum − (sum of the f+1 largest): one divergence-free selection
+    kernel pass over the K×K matrix instead of a host topk (the round-1
+    bottleneck at K=1e4).
+    """
+    n = sqdists.shape[0]
+    if n - f - 2 < 1:
+        raise ValueError(f"krum needs n - f - 2 >= 1 (n={n}, f={f})")
+    ext = _route(sqdists)
+    if ext is not None:
+        D = sqdists
+        if bool((D.diagonal() != 0).any()):
+            D = D.clone()
+            D.fill_diagonal_(0.0)
+        return ext.col_trimmed_sum(_prep(D), 0, f + 1)
+    return _ref.krum_scores(sqdists, f)
 
 
 def trimmed_mean(U: Tensor, b: int) -> Tensor:
@@ -209,6 +245,7 @@ __all__ = [
     "weighted_col_sum", "masked_col_mean", "masked_col_mean_std",
     "row_sq_norms", "row_norms", "row_diff_norms", "row_dots",
     "pairwise_sq_dists", "gram", "cos_sim_gram",
-    "krum_scores", "clip_to_norm", "centered_clip_iter", "row_clip_to_norm_",
+    "krum_scores", "col_trimmed_sum",
+    "clip_to_norm", "centered_clip_iter", "row_clip_to_norm_",
     "philox_normal", "flat_sgd_step_", "sanitize_", "hip_available",
 ]

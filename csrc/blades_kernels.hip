@@ -301,6 +301,152 @@ __global__ void trimmed_select_kernel(const float* __restrict__ U,
 }
 
 // ---------------------------------------------------------------------------
+// K3/K2 small-b fast path — divergence-free register selection
+// ---------------------------------------------------------------------------
+// The LDS-heap kernel above is divergence-bound at the headline shape
+// (K=100, b=20): with 64 lanes and b/K = 0.2, some lane replaces on nearly
+// every row, so the ballot-gated sift serializes the wave ~every step
+// (measured 7.5 ms vs the 0.7 ms HBM floor; PMC showed SQ busy ~12%,
+// profiles/r01_pmc_aggregation.csv).  This kernel has NO data-dependent
+// control flow in the hot path at all:
+//
+//   * one coordinate per LANE; each lane keeps the NB smallest (lo, sorted
+//     ascending) and NB largest (hi, sorted descending) values seen, in
+//     REGISTERS with static indices (b <= NB <= 32);
+//   * rows stream in groups of NB: one in-register bitonic sort of the
+//     group (min/max compare-exchange network, fully unrolled), then a
+//     bitonic lower-half merge lo[i] <-> g[NB-1-i] + 5-stage clean — every
+//     lane executes the identical instruction stream;
+//   * a wave ballot gates the sort+merge on "any lane's group crosses its
+//     thresholds": always true in the first rows, vanishingly rare once
+//     K >> b (expected triggered groups ~ 64·NB·ln(K)/K of rows), so the
+//     kernel degrades to a pure float-stream + f64 accumulate at large K;
+//   * fp64 running sum (cancellation under Byzantine-magnitude outliers:
+//     result = sum − trims; see commit 7977a4c).
+//
+// VALU cost per value when every group triggers (K ~ 100): bitonic sort
+// ~240 CE + 2 merges ~112 CE each = ~930 min/max ops per NB=32 group
+// = ~29/value -> ~0.45 ms of VALU at K=100, d=11.2M — under the 0.71 ms
+// HBM floor, i.e. memory-bound by construction.
+
+template <int NB>
+__device__ __forceinline__ void bitonic_sort_asc(float (&g)[NB]) {
+  #pragma unroll
+  for (int k = 2; k <= NB; k <<= 1) {
+    #pragma unroll
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      #pragma unroll
+      for (int i = 0; i < NB; ++i) {
+        const int l = i ^ j;
+        if (l > i) {
+          const bool up = ((i & k) == 0);
+          const float a = g[i], b = g[l];
+          g[i] = up ? fminf(a, b) : fmaxf(a, b);
+          g[l] = up ? fmaxf(a, b) : fminf(a, b);
+        }
+      }
+    }
+  }
+}
+
+// clean a bitonic sequence into ascending (ASC) or descending order
+template <int NB, bool ASC>
+__device__ __forceinline__ void bitonic_clean(float (&m)[NB]) {
+  #pragma unroll
+  for (int j = NB >> 1; j > 0; j >>= 1) {
+    #pragma unroll
+    for (int i = 0; i < NB; ++i) {
+      const int l = i ^ j;
+      if (l > i) {
+        const float a = m[i], b = m[l];
+        m[i] = ASC ? fminf(a, b) : fmaxf(a, b);
+        m[l] = ASC ? fmaxf(a, b) : fminf(a, b);
+      }
+    }
+  }
+}
+
+// b_lo / b_hi: drop that many smallest / largest values (either may be 0;
+// both <= NB).  inv scales the trimmed sum: 1/(K-b_lo-b_hi) for the
+// trimmed mean, 1.0 for raw trimmed sums (K5 krum scores: the score of
+// column i of the symmetric zero-diagonal distance matrix is the sum of
+// its n-f-2 smallest off-diagonal entries = col_trimmed_sum(D, 0, f+1),
+// reference: aggregators/krum.py:9-25).
+template <int NB>
+__global__ __launch_bounds__(256)
+void trimmed_regsel_kernel(const float* __restrict__ U,
+                           float* __restrict__ out,
+                           long long K, long long d, long long ld,
+                           int b_lo, int b_hi, double inv) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long nthreads = (long long)gridDim.x * blockDim.x;
+
+  for (long long j = tid; j < d; j += nthreads) {
+    float lo[NB], hi[NB], g[NB];
+    #pragma unroll
+    for (int i = 0; i < NB; ++i) { lo[i] = INFINITY; hi[i] = -INFINITY; }
+    // 4 partial f64 accumulators break the serial add chain in the
+    // unrolled group loop (dependent v_add_f64 latency would serialize)
+    double s0 = 0.0, s1 = 0.0, s2 = 0.0, s3 = 0.0;
+
+    long long k = 0;
+    for (; k + NB <= K; k += NB) {
+      // group load + f64 accumulate + threshold test (branch-free)
+      bool need = false;
+      #pragma unroll
+      for (int u = 0; u < NB; ++u) {
+        const float v = U[(k + u) * ld + j];
+        g[u] = v;
+        if ((u & 3) == 0) s0 += (double)v;
+        else if ((u & 3) == 1) s1 += (double)v;
+        else if ((u & 3) == 2) s2 += (double)v;
+        else s3 += (double)v;
+        need |= (v < lo[NB - 1]) | (v > hi[NB - 1]);
+      }
+      if (__any(need)) {
+        bitonic_sort_asc<NB>(g);
+        // lower-half bitonic merges IN PLACE (lo/hi are overwritten by
+        // the merge result anyway — no temporaries, ~96 regs peak)
+        #pragma unroll
+        for (int i = 0; i < NB; ++i) lo[i] = fminf(lo[i], g[NB - 1 - i]);
+        bitonic_clean<NB, true>(lo);
+        #pragma unroll
+        for (int i = 0; i < NB; ++i) hi[i] = fmaxf(hi[i], g[NB - 1 - i]);
+        bitonic_clean<NB, false>(hi);
+      }
+    }
+    // tail rows: branchless bubble insert (static indices)
+    for (; k < K; ++k) {
+      const float v = U[k * ld + j];
+      s0 += (double)v;
+      float c = v;
+      #pragma unroll
+      for (int i = 0; i < NB; ++i) {
+        const float a = lo[i];
+        lo[i] = fminf(a, c);
+        c = fmaxf(a, c);
+      }
+      c = v;
+      #pragma unroll
+      for (int i = 0; i < NB; ++i) {
+        const float a = hi[i];
+        hi[i] = fmaxf(a, c);
+        c = fminf(a, c);
+      }
+    }
+    // predicated static-index sum — a runtime-bound loop would force the
+    // register arrays to scratch (dynamic indexing)
+    double trim = 0.0;
+    #pragma unroll
+    for (int i = 0; i < NB; ++i) {
+      trim += (i < b_lo) ? (double)lo[i] : 0.0;
+      trim += (i < b_hi) ? (double)hi[i] : 0.0;
+    }
+    out[j] = (float)(((s0 + s1) + (s2 + s3) - trim) * inv);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // K2/K3 at large K — dual radix-select trimmed mean
 // ---------------------------------------------------------------------------
 // The LDS streaming-selection kernel above needs 2·b floats of LDS per
@@ -328,7 +474,8 @@ constexpr int RT_T = 32;  // coordinates per block tile
 __global__ __launch_bounds__(256)
 void radix_trimmed_kernel(const float* __restrict__ U,
                           float* __restrict__ out,
-                          long long K, long long d, long long ld, long b) {
+                          long long K, long long d, long long ld,
+                          long b_lo, long b_hi, double inv) {
   __shared__ unsigned int histA[RT_T][256];
   __shared__ unsigned int histB[RT_T][256];
   __shared__ unsigned int prefA[RT_T], baseA[RT_T];
@@ -336,8 +483,8 @@ void radix_trimmed_kernel(const float* __restrict__ U,
   __shared__ double fsum[RT_T];            // final-pass per-coord sums
   __shared__ unsigned int feqA[RT_T], feqB[RT_T];
 
-  const long long kA = b;             // first kept rank (0-indexed)
-  const long long kB = K - b - 1;     // last kept rank
+  const long long kA = b_lo;          // first kept rank (0-indexed)
+  const long long kB = K - b_hi - 1;  // last kept rank
 
   for (long long j0 = (long long)blockIdx.x * RT_T; j0 < d;
        j0 += (long long)gridDim.x * RT_T) {
@@ -470,16 +617,16 @@ void radix_trimmed_kernel(const float* __restrict__ U,
       const long long cltA = baseA[t], cltB = baseB[t];
       double total;
       if (prefA[t] == prefB[t]) {
-        total = (double)(K - 2 * b) * vA;
+        total = (double)(K - b_lo - b_hi) * vA;
       } else {
-        // kept ranks are [b, K-b); ties at the thresholds contribute the
-        // overlap of their rank range with the kept band
+        // kept ranks are [b_lo, K-b_hi); ties at the thresholds contribute
+        // the overlap of their rank range with the kept band
         const long long incA =
-            (cltA + ea < K - b ? cltA + ea : K - b) - b;
-        const long long incB = (K - b) - (cltB > b ? cltB : b);
+            (cltA + ea < K - b_hi ? cltA + ea : K - b_hi) - b_lo;
+        const long long incB = (K - b_hi) - (cltB > b_lo ? cltB : b_lo);
         total = s + (double)incA * vA + (double)incB * vB;
       }
-      out[j0 + t] = (float)(total / (double)(K - 2 * b));
+      out[j0 + t] = (float)(total * inv);
     }
     __syncthreads();
   }
@@ -739,30 +886,66 @@ masked_col_mean_std(torch::Tensor U, torch::Tensor mask, bool unbiased,
   return {mu, sd};
 }
 
-torch::Tensor trimmed_mean_select(torch::Tensor U, long b) {
+// General asymmetric trimmed column sum: drop b_lo smallest + b_hi
+// largest per column, return inv * (sum of the rest).
+static torch::Tensor trimmed_core(torch::Tensor U, long b_lo, long b_hi,
+                                  double inv) {
   auto v = view_of(U);
-  TORCH_CHECK(v.K - 2 * b >= 1, "trimmed_mean needs K > 2b");
-  TORCH_CHECK(b >= 0, "b must be >= 0");
+  TORCH_CHECK(b_lo >= 0 && b_hi >= 0, "trims must be >= 0");
+  TORCH_CHECK(v.K - b_lo - b_hi >= 1, "trimmed sum needs K > b_lo + b_hi");
   auto out = torch::empty({v.d}, U.options());
   auto stream = c10::hip::getCurrentHIPStream().stream();
-  // pick block size so the 2*b per-thread LDS floats fit in 64 KiB
-  int BS = 256;
-  size_t need = 2 * (size_t)b * 4 * BS;
-  while (BS > 64 && need > 64 * 1024) { BS /= 2; need /= 2; }
-  if (need <= 64 * 1024 && b > 0) {
+  const long bmax = std::max(b_lo, b_hi);
+  // small trims: divergence-free register-selection kernel (headline
+  // path).  NB = smallest capacity >= bmax so a b=2 median does not pay
+  // the NB=32 sort network.
+  if (bmax <= 32) {
+    const int BS = 256;
     const int grid = col_grid(v.d, BS);
-    trimmed_select_kernel<<<grid, BS, need, stream>>>(
-        v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b);
+    if (bmax <= 8)
+      trimmed_regsel_kernel<8><<<grid, BS, 0, stream>>>(
+          v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo,
+          (int)b_hi, inv);
+    else if (bmax <= 16)
+      trimmed_regsel_kernel<16><<<grid, BS, 0, stream>>>(
+          v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo,
+          (int)b_hi, inv);
+    else
+      trimmed_regsel_kernel<32><<<grid, BS, 0, stream>>>(
+          v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo,
+          (int)b_hi, inv);
     return out;
   }
-  if (b == 0) {
-    return col_mean(U) ;
+  // mid symmetric b: LDS-heap streaming selection (2*b per-thread LDS)
+  if (b_lo == b_hi && inv != 1.0) {
+    int BS = 256;
+    size_t need = 2 * (size_t)b_lo * 4 * BS;
+    while (BS > 64 && need > 64 * 1024) { BS /= 2; need /= 2; }
+    if (need <= 64 * 1024) {
+      const int grid = col_grid(v.d, BS);
+      trimmed_select_kernel<<<grid, BS, need, stream>>>(
+          v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo);
+      return out;
+    }
   }
-  // large b (median at K >= ~260): dual radix-select kernel
+  // large / asymmetric trims: dual radix-select kernel
   const int grid = (int)std::min<long long>(cdiv(v.d, RT_T), kMaxBlocks);
   radix_trimmed_kernel<<<grid, 256, 0, stream>>>(
-      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b);
+      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b_lo, b_hi, inv);
   return out;
+}
+
+torch::Tensor trimmed_mean_select(torch::Tensor U, long b) {
+  TORCH_CHECK(U.size(0) - 2 * b >= 1, "trimmed_mean needs K > 2b");
+  TORCH_CHECK(b >= 0, "b must be >= 0");
+  if (b == 0) {
+    return col_mean(U);
+  }
+  return trimmed_core(U, b, b, 1.0 / (double)(U.size(0) - 2 * b));
+}
+
+torch::Tensor col_trimmed_sum(torch::Tensor U, long b_lo, long b_hi) {
+  return trimmed_core(U, b_lo, b_hi, 1.0);
 }
 
 torch::Tensor trimmed_mean(torch::Tensor U, long b) {
@@ -770,7 +953,7 @@ torch::Tensor trimmed_mean(torch::Tensor U, long b) {
 }
 
 torch::Tensor trimmed_mean_radix(torch::Tensor U, long b) {
-  // direct radix path (A/B benchmarking; auto-dispatch uses the LDS
+  // direct radix path (A/B benchmarking; auto-dispatch uses the register
   // selection kernel for small b)
   auto v = view_of(U);
   TORCH_CHECK(v.K - 2 * b >= 1 && b > 0, "bad b");
@@ -778,7 +961,8 @@ torch::Tensor trimmed_mean_radix(torch::Tensor U, long b) {
   auto stream = c10::hip::getCurrentHIPStream().stream();
   const int grid = (int)std::min<long long>(cdiv(v.d, RT_T), kMaxBlocks);
   radix_trimmed_kernel<<<grid, 256, 0, stream>>>(
-      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b);
+      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b, b,
+      1.0 / (double)(v.K - 2 * b));
   return out;
 }
 
@@ -862,6 +1046,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "masked column mean+std (K10)", py::arg("U"), py::arg("mask"),
         py::arg("unbiased") = true, py::arg("count") = -1.0);
   m.def("trimmed_mean", &trimmed_mean, "coordinate-wise trimmed mean (K3)");
+  m.def("col_trimmed_sum", &col_trimmed_sum,
+        "asymmetric trimmed column sum (K5 krum scores)");
   m.def("trimmed_mean_radix", &trimmed_mean_radix,
         "trimmed mean via dual radix select (benchmarking entry)");
   m.def("col_median", &col_median, "coordinate-wise median (K2)");

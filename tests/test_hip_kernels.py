@@ -68,7 +68,9 @@ def test_masked_col_mean_std(ext, K, d):
     assert torch.allclose(m2, rmu, atol=1e-5, rtol=1e-5)
 
 
-@pytest.mark.parametrize("K,b", [(9, 2), (100, 20), (100, 49), (64, 31), (7, 0)])
+@pytest.mark.parametrize("K,b", [(9, 2), (100, 20), (100, 49), (64, 31),
+                                 (7, 0), (20, 8), (70, 32), (33, 16),
+                                 (1000, 20), (12, 5)])
 def test_trimmed_mean(ext, K, b):
     from blades_amd.ops import torch_ref
     U = randU(K, 50000, seed=b)
@@ -125,6 +127,36 @@ def test_pairwise_sq_dists_vs_ref(ext):
     ref = torch_ref.pairwise_sq_dists(U)
     assert torch.allclose(D, ref, rtol=1e-3, atol=1e-1)
     assert (D.diagonal() == 0).all()
+
+
+@pytest.mark.parametrize("K,f", [(25, 5), (100, 20), (1000, 10),
+                                 (1000, 200), (2000, 660)])
+def test_krum_scores_kernel(ext, K, f):
+    """K5 — the selection-kernel scores must match the host topk reference
+    on the symmetric zero-diagonal distance matrix (small f -> register
+    kernel, large f -> asymmetric radix path)."""
+    from blades_amd import ops
+    from blades_amd.ops import torch_ref
+    U = randU(K, 256, seed=K + f)
+    D = ops.pairwise_sq_dists(U)
+    s_gpu = ops.krum_scores(D, f)
+    s_ref = torch_ref.krum_scores(D.double(), f).float()
+    assert torch.allclose(s_gpu, s_ref, rtol=1e-5, atol=1e-3), \
+        (s_gpu - s_ref).abs().max().item()
+
+
+@pytest.mark.parametrize("K,blo,bhi", [(50, 0, 21), (50, 3, 0), (100, 5, 40),
+                                       (300, 0, 120)])
+def test_col_trimmed_sum_asymmetric(ext, K, blo, bhi):
+    from blades_amd import ops
+    U = randU(K, 30000, seed=K + blo + bhi)
+    out = ops.col_trimmed_sum(U, blo, bhi)
+    s = U.double().sum(0)
+    if blo:
+        s -= torch.topk(U.double(), blo, dim=0, largest=False).values.sum(0)
+    if bhi:
+        s -= torch.topk(U.double(), bhi, dim=0, largest=True).values.sum(0)
+    assert torch.allclose(out, s.float(), rtol=1e-5, atol=1e-4)
 
 
 def test_centered_clip_iter(ext):

@@ -83,3 +83,26 @@ def test_single_outlier_bounded_influence(K, d, seed):
         out = agg(U_attacked.clone())
         assert (out >= lo - 1e-4).all() and (out <= hi + 1e-4).all(), \
             type(agg).__name__
+
+
+def test_col_trimmed_sum_cpu_fallback():
+    import torch
+
+    from blades_amd import ops
+    U = torch.randn(40, 100)
+    out = ops.col_trimmed_sum(U, 3, 7)
+    s = U.double().sum(0)
+    s -= torch.topk(U.double(), 3, dim=0, largest=False).values.sum(0)
+    s -= torch.topk(U.double(), 7, dim=0, largest=True).values.sum(0)
+    assert torch.allclose(out, s.float(), atol=1e-5)
+
+
+def test_krum_scores_cpu_matches_ref():
+    import torch
+
+    from blades_amd import ops
+    from blades_amd.ops import torch_ref
+    U = torch.randn(30, 64)
+    D = torch_ref.pairwise_sq_dists(U)
+    assert torch.allclose(ops.krum_scores(D, 5), torch_ref.krum_scores(D, 5),
+                          atol=1e-4)
