@@ -33,9 +33,13 @@ class SyntheticFLDataset:
         self.device = torch.device(device)
         self.test_samples_per_client = test_samples_per_client
         # learnable=True labels samples with a fixed random linear teacher
-        # (same teacher for train and test), so eval loss can actually drop
+        # (same teacher for train and test), so eval loss can actually drop;
+        # learnable="templates" draws X = template[y] + 0.5*noise from
+        # num_classes fixed random templates — quickly learnable at conv
+        # scale (the robustness-curve demonstration)
         self.learnable = learnable
         self._teacher = None
+        self._templates = None
         self._clients = list(range(num_clients))
         self._cursor = [0] * num_clients
         # lazy pools: client -> (X [S,*shape], y [S])
@@ -54,7 +58,15 @@ class SyntheticFLDataset:
         g = torch.Generator()
         g.manual_seed(client_philox_seed(self.seed, int(u_id), 0, tag=tag))
         X = torch.randn((n, *self.shape), generator=g)
-        if self.learnable:
+        if self.learnable == "templates":
+            if self._templates is None:
+                gt = torch.Generator()
+                gt.manual_seed(client_philox_seed(self.seed, 0, 0, tag=98))
+                self._templates = torch.randn((self.num_classes, *self.shape),
+                                              generator=gt)
+            y = torch.randint(0, self.num_classes, (n,), generator=g)
+            X = self._templates[y] + 0.5 * X
+        elif self.learnable:
             if self._teacher is None:
                 gt = torch.Generator()
                 gt.manual_seed(client_philox_seed(self.seed, 0, 0, tag=99))

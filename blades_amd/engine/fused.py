@@ -215,8 +215,15 @@ class FusedEngine:
             U.mul_(-lr)
             return U
 
-        # general FedAvg path: per-client divergent weights in a [C, d] slab
-        slab = theta.unsqueeze(0).repeat(C, 1).contiguous()
+        # general FedAvg path: per-client divergent weights in a [C, d]
+        # slab.  The buffer persists across rounds (the per-round
+        # `theta.repeat` allocation was ~25% of the config-3 step,
+        # VERDICT r1 weak #6); only the broadcast-copy of θ remains.
+        if (getattr(self, "_slab", None) is None
+                or self._slab.shape[0] != C):
+            self._slab = torch.empty(C, self.spec.d, device=self.device)
+        slab = self._slab
+        slab.copy_(theta.unsqueeze(0).expand(C, -1))
         slab_views = self.spec.batched_views(slab)
         grad_fn = self._build_grad_fn(shared_params=False)
         for s in range(local_steps):
@@ -282,7 +289,11 @@ class FusedEngine:
             U.mul_(-lr)
             return U
 
-        slab = theta.unsqueeze(0).repeat(C, 1).contiguous()
+        if (getattr(self, "_slab", None) is None
+                or self._slab.shape[0] != C):
+            self._slab = torch.empty(C, self.spec.d, device=self.device)
+        slab = self._slab
+        slab.copy_(theta.unsqueeze(0).expand(C, -1))
         slab_views = self.spec.batched_views(slab)
         for s in range(local_steps):
             params = {n: slab_views[n].detach().requires_grad_()
