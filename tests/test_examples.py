@@ -37,3 +37,13 @@ def test_cli_driver_runs(tmp_path):
     )
     assert proc.returncode == 0, proc.stderr[-2000:]
     assert "finished 2 rounds" in proc.stdout
+
+
+@pytest.mark.timeout(1200)
+def test_docs_build_executes_gallery(tmp_path):
+    """docs/build.py is the docs-as-CI gate (the reference executed its
+    examples through sphinx-gallery, docs/source/conf.py:75-79)."""
+    p = subprocess.run([sys.executable, os.path.join(REPO, "docs", "build.py")],
+                       capture_output=True, text=True, cwd=REPO, timeout=1100)
+    assert p.returncode == 0, p.stdout + p.stderr
+    assert os.path.exists(os.path.join(REPO, "docs", "gallery", "index.md"))
