@@ -405,13 +405,17 @@ void trimmed_regsel_kernel(const float* __restrict__ U,
       }
       if (__any(need)) {
         bitonic_sort_asc<NB>(g);
-        // lower-half bitonic merges IN PLACE (lo/hi are overwritten by
-        // the merge result anyway — no temporaries, ~96 regs peak)
+        // half bitonic merges IN PLACE (lo/hi are overwritten by the
+        // merge result anyway — no temporaries, ~96 regs peak).
+        // lo (ascending) pairs with reversed g; hi (descending) pairs
+        // with g directly: by the negation duality (-hi ascending,
+        // reversed(-g) ascending) the lower-half cleaner min(-hi[i],
+        // -g[NB-1-(NB-1-i)]) = -max(hi[i], g[i]).
         #pragma unroll
         for (int i = 0; i < NB; ++i) lo[i] = fminf(lo[i], g[NB - 1 - i]);
         bitonic_clean<NB, true>(lo);
         #pragma unroll
-        for (int i = 0; i < NB; ++i) hi[i] = fmaxf(hi[i], g[NB - 1 - i]);
+        for (int i = 0; i < NB; ++i) hi[i] = fmaxf(hi[i], g[i]);
         bitonic_clean<NB, false>(hi);
       }
     }

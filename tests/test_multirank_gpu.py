@@ -2,17 +2,18 @@
 
 RCCL refuses two ranks on one device ("Duplicate GPU detected", probed in
 round 2 — gpurun_out/r2_gputest1.log) and CPX partitioning is blocked on
-this pool, so a single-GPU lease cannot run a true 2-rank RCCL job.  What
-CAN be hardware-proven here:
+this pool (gpurun_out/cpx_probe.log), so a single-GPU lease cannot run a
+true 2-rank RCCL job.  What IS hardware-proven here:
 
-* the RCCL backend executes every collective the runtime uses at ws=1,
-  inside hipGraph capture (validates RCCL stream-capture support + the
-  exact dist calls the multi-rank CapturedRound records);
+* the RCCL backend executes every collective the runtime uses at ws=1
+  INSIDE hipGraph capture+replay (scripts/probe_rccl_graph.py — validates
+  RCCL stream-capture support, the mechanism multi-rank CapturedRound
+  relies on);
 * a REAL 2-process round with all client math on the GPU, collectives
   staged over gloo (BLADES_AMD_BACKEND=gloo) — the full multi-rank round
   logic (shard split, slab layout, all-gather/all-to-all re-shard,
   rank-local attacks, shard aggregation) runs on hardware and must
-  reproduce the single-rank θ bitwise.
+  reproduce the single-rank θ.
 """
 import os
 import subprocess
@@ -20,109 +21,72 @@ import sys
 
 import pytest
 import torch
+import torch.multiprocessing as mp
 
 pytestmark = pytest.mark.gpu
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def _run(script: str, env=None, nproc=None, timeout=240):
-    e = dict(os.environ)
-    if env:
-        e.update(env)
-    if nproc:
-        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-               f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
-               "--master-port", "29655", "-"]
-    else:
-        cmd = [sys.executable, "-"]
-    p = subprocess.run(cmd, input=script, text=True, capture_output=True,
-                       env=e, cwd=REPO, timeout=timeout)
-    assert p.returncode == 0, f"stdout:\n{p.stdout}\nstderr:\n{p.stderr}"
-    return p.stdout
-
-
 @pytest.mark.timeout(300)
 def test_rccl_collectives_inside_hipgraph_ws1():
-    """Every runtime collective through RCCL, captured in a hipGraph and
-    replayed — the mechanism multi-rank CapturedRound relies on."""
-    _run(r"""
-import torch, torch.distributed as dist
-dist.init_process_group("nccl", init_method="tcp://127.0.0.1:29656",
-                        world_size=1, rank=0)
-dev = torch.device("cuda:0")
-torch.cuda.set_device(dev)
-send = torch.ones(1 << 20, device=dev)
-gath = torch.empty(1 << 20, device=dev)
-a2a_s = torch.arange(1024, dtype=torch.float32, device=dev)
-a2a_r = torch.empty_like(a2a_s)
-red = torch.full((4096,), 2.0, device=dev)
-# warm the communicator outside capture (comm init is not capturable)
-dist.all_gather_into_tensor(gath, send)
-dist.all_to_all_single(a2a_r, a2a_s)
-dist.all_reduce(red)
-dist.broadcast(red, src=0)
-torch.cuda.synchronize()
-send.fill_(3.0); a2a_s.fill_(7.0); red.fill_(5.0)
-g = torch.cuda.CUDAGraph()
-with torch.cuda.graph(g):
-    dist.all_gather_into_tensor(gath, send)
-    dist.all_to_all_single(a2a_r, a2a_s)
-    dist.all_reduce(red)
-gath.zero_(); a2a_r.zero_(); torch.cuda.synchronize()
-g.replay(); torch.cuda.synchronize()
-assert gath.mean().item() == 3.0, gath.mean().item()
-assert a2a_r.mean().item() == 7.0
-assert red[0].item() == 5.0
-# replay again with new inputs through the static buffers
-send.fill_(9.0)
-g.replay(); torch.cuda.synchronize()
-assert gath.mean().item() == 9.0
-dist.destroy_process_group()
-print("RCCL-in-hipGraph OK")
-""")
+    p = subprocess.run(
+        [sys.executable, os.path.join(REPO, "scripts", "probe_rccl_graph.py")],
+        capture_output=True, text=True, cwd=REPO, timeout=240)
+    assert p.returncode == 0, f"stdout:\n{p.stdout}\nstderr:\n{p.stderr[-3000:]}"
+    assert "step8 capture+replay a2a/allreduce ok" in p.stdout
 
 
-def _round_script(gather: str, tag: str) -> str:
-    return f"""
-import os, torch
-os.environ.setdefault("LOCAL_RANK", "0")   # both ranks share cuda:0
-from blades_amd import Simulator
-from blades_amd.datasets import SyntheticFLDataset
-from blades_amd.models import MLP
+def _gpu_round(rank, world, port, gather, out_q, seed=5):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(world), "LOCAL_RANK": "0",
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "BLADES_AMD_BACKEND": "gloo",
+    })
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import MLP
 
-rank = int(os.environ.get("RANK", "0"))
-ds = SyntheticFLDataset(num_clients=6, samples_per_client=16, batch_size=8,
-                        shape=(1, 28, 28), num_classes=10, seed=0,
-                        device="cuda:0")
-sim = Simulator(ds, num_byzantine=2, attack="alie",
-                attack_kws={{"num_clients": 6, "num_byzantine": 2}},
-                aggregator="trimmedmean", aggregator_kws={{"nb": 2}},
-                use_cuda=True, device="cuda:0", gather="{gather}",
-                log_path=f"/tmp/bl_mrgpu_{tag}_{{rank}}", seed=5)
-torch.manual_seed(5)
-sim.run(MLP(), global_rounds=3, validate_interval=0, client_lr=0.1,
-        server_lr=1.0)
-theta = sim.server.flat_parameters().cpu()
-torch.save(theta, f"/tmp/bl_mrgpu_{tag}_{{rank}}.pt")
-print("rank", rank, "done", float(theta.abs().sum()))
-"""
+    ds = SyntheticFLDataset(num_clients=6, samples_per_client=16,
+                            batch_size=8, shape=(1, 28, 28), num_classes=10,
+                            seed=0, device="cuda:0")
+    sim = Simulator(ds, num_byzantine=2, attack="alie",
+                    attack_kws={"num_clients": 6, "num_byzantine": 2},
+                    aggregator="trimmedmean", aggregator_kws={"nb": 2},
+                    use_cuda=True, device="cuda:0", gather=gather,
+                    log_path=f"/tmp/bl_mrgpu_{gather}_{world}_{rank}",
+                    seed=seed)
+    torch.manual_seed(seed)
+    sim.run(MLP(), global_rounds=3, validate_interval=0, client_lr=0.1,
+            server_lr=1.0)
+    out_q.put((rank, sim.server.flat_parameters().cpu().numpy().copy()))
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
 
 
-@pytest.mark.timeout(600)
+def _spawn_world(world, port, gather):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_gpu_round, args=(r, world, port, gather, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    thetas = {}
+    for _ in range(world):
+        rank, theta = q.get(timeout=400)
+        thetas[rank] = torch.from_numpy(theta)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    return thetas
+
+
+@pytest.mark.timeout(900)
 @pytest.mark.parametrize("gather", ["full", "coordinate"])
 def test_two_process_gpu_round_matches_single_rank(gather):
-    tag1 = f"ws1_{gather}"
-    _run(_round_script(gather="full", tag=tag1))  # ws=1: gather is moot
-    theta1 = torch.load(f"/tmp/bl_mrgpu_{tag1}_0.pt", weights_only=True)
-
-    tag2 = f"ws2_{gather}"
-    _run(_round_script(gather=gather, tag=tag2),
-         env={"BLADES_AMD_BACKEND": "gloo",
-              "HSA_ENABLE_IPC_MODE_LEGACY": "0"},
-         nproc=2, timeout=480)
-    t0 = torch.load(f"/tmp/bl_mrgpu_{tag2}_0.pt", weights_only=True)
-    t1 = torch.load(f"/tmp/bl_mrgpu_{tag2}_1.pt", weights_only=True)
-    assert torch.equal(t0, t1), "ranks diverged"
-    assert torch.allclose(theta1, t0, atol=1e-6), \
-        (theta1 - t0).abs().max().item()
+    theta1 = _spawn_world(1, 29668, "full")[0]
+    thetas = _spawn_world(2, 29669, gather)
+    assert torch.equal(thetas[0], thetas[1]), "ranks diverged"
+    assert torch.allclose(theta1, thetas[0], atol=1e-6), \
+        (theta1 - thetas[0]).abs().max().item()
