@@ -209,10 +209,11 @@ class FusedEngine:
             U = out if out is not None else torch.empty(
                 (C, self.spec.d), device=self.device)
             views = self.spec.batched_views(U)
+            neg_lr = -lr if not isinstance(lr, torch.Tensor) else lr.neg()
             with torch.no_grad():
+                # fused scale+store: one slab pass instead of copy + mul_
                 for name, g in zip(self.spec.names, grads):
-                    views[name].copy_(g)
-            U.mul_(-lr)
+                    torch.mul(g, neg_lr, out=views[name])
             return U
 
         # general FedAvg path: per-client divergent weights in a [C, d]
