@@ -29,12 +29,25 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 @pytest.mark.timeout(300)
-def test_rccl_collectives_inside_hipgraph_ws1():
-    p = subprocess.run(
-        [sys.executable, os.path.join(REPO, "scripts", "probe_rccl_graph.py")],
-        capture_output=True, text=True, cwd=REPO, timeout=240)
-    assert p.returncode == 0, f"stdout:\n{p.stdout}\nstderr:\n{p.stderr[-3000:]}"
-    assert "step8 capture+replay a2a/allreduce ok" in p.stdout
+def test_rccl_collectives_inside_hipgraph_ws1(tmp_path):
+    # file-backed output (not pipes) + explicit kill so a hang shows its
+    # last completed step instead of a bare TimeoutExpired
+    log = tmp_path / "probe.log"
+    with open(log, "w") as f:
+        p = subprocess.Popen(
+            [sys.executable, "-u",
+             os.path.join(REPO, "scripts", "probe_rccl_graph.py")],
+            stdout=f, stderr=subprocess.STDOUT, cwd=REPO,
+            stdin=subprocess.DEVNULL)
+        try:
+            rc = p.wait(timeout=240)
+        except subprocess.TimeoutExpired:
+            p.kill()
+            p.wait(timeout=30)
+            rc = None
+    out = open(log).read()
+    assert rc == 0, f"probe rc={rc}; output:\n{out[-4000:]}"
+    assert "step8 capture+replay a2a/allreduce ok" in out
 
 
 def _gpu_round(rank, world, port, gather, out_q, seed=5):
