@@ -137,3 +137,53 @@ def test_transforms_shapes():
     assert out.shape == (3, 32, 32)
     xb = torch.rand(5, 3, 32, 32)
     assert t(xb).shape == (5, 3, 32, 32)
+
+
+def test_leaf_cli_pipeline(tmp_path):
+    """Full LEAF preprocessing pipeline on a tiny synthetic all_data set
+    (reference: models/utils/preprocess.sh stages)."""
+    import json
+    import subprocess
+    import sys
+
+    from blades_amd.datasets import leaf_cli
+
+    data_dir = tmp_path / "femnist" / "data"
+    all_dir = data_dir / "all_data"
+    all_dir.mkdir(parents=True)
+    users = [f"u{i}" for i in range(20)]
+    data = {u: {"x": [[float(i), 0.5]] * (i + 1), "y": [i % 3] * (i + 1)}
+            for i, u in enumerate(users)}
+    leaf_cli.write_leaf(str(all_dir / "all_data_0.json"), users, data)
+
+    leaf_cli.preprocess(str(data_dir), sample_mode="niid", fraction=0.5,
+                        min_samples=3, train_fraction=0.8)
+    tr_users, tr_counts, tr_data = leaf_cli.load_dir(str(data_dir / "train"))
+    te_users, te_counts, te_data = leaf_cli.load_dir(str(data_dir / "test"))
+    assert tr_users and te_users
+    assert set(te_users) <= set(tr_users)  # sample-split keeps users
+    for u, c in zip(tr_users, tr_counts):
+        assert c >= 1
+        # x/y lengths consistent
+        assert len(tr_data[u]["x"]) == len(tr_data[u]["y"]) == c
+    # checksum manifest written and verifies
+    manifest = data_dir / "meta" / "dir-checksum.md5.json"
+    assert manifest.exists()
+    assert leaf_cli.checksum(str(data_dir), str(manifest), verify=True)
+    # seeds persisted (preprocess.sh meta behavior)
+    assert (data_dir / "meta" / "sampling_seed.txt").exists()
+
+    # stats + CLI entry
+    out = leaf_cli.stats(str(data_dir / "train"))
+    assert "users:" in out
+    p = subprocess.run([sys.executable, "-m",
+                        "blades_amd.datasets.leaf_cli", "stats",
+                        "--src", str(all_dir)],
+                       capture_output=True, text=True)
+    assert p.returncode == 0 and "users: 20" in p.stdout
+
+    # iid sampling path
+    leaf_cli.sample(str(all_dir), str(tmp_path / "iid"), 0.3, iid=True,
+                    iid_user_fraction=0.2, seed=1)
+    iu, ic, _ = leaf_cli.load_dir(str(tmp_path / "iid"))
+    assert len(iu) == 4 and sum(ic) == int(0.3 * sum(i + 1 for i in range(20)))
