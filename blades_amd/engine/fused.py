@@ -67,7 +67,6 @@ class FusedEngine:
     def __init__(self, model: nn.Module, spec: ParamSpec, device,
                  client_chunk: Optional[int] = None):
         import copy
-        import os
 
         self.device = torch.device(device)
         self.spec = spec
@@ -76,22 +75,9 @@ class FusedEngine:
         # buffers (none for vmap-safe BN; kept for generality) are shared
         self.buffers = {k: v.detach().clone() for k, v in self.base.named_buffers()}
         self.client_chunk = client_chunk
-        # population path: direct MFMA popconv kernels for supported models.
-        # Measured round 1: the popconv round runs ~467 ms vs 229 ms for the
-        # vmap+hipGraph path on the headline config (popconv_fwd reaches
-        # 57-60 TF f32 = 37% of MFMA peak, but padded-plane FLOP inflation
-        # on the deep layers plus the unfused BN/elementwise chain eat the
-        # win) — so it is OPT-IN (BLADES_AMD_POPCONV=1 / FORCE_POP for CPU
-        # parity tests) until the kernel is pipelined and BN is fused.
-        from blades_amd.engine.popmodel import population_forward_for
-
-        self._pop_fn = None
-        want_pop = (os.environ.get("BLADES_AMD_POPCONV") == "1"
-                    or os.environ.get("BLADES_AMD_FORCE_POP") == "1")
-        if want_pop and os.environ.get("BLADES_AMD_NO_POPCONV", "0") != "1":
-            fn = population_forward_for(model)
-            if fn is not None:
-                self._pop_fn = fn
+        # NOTE round 2: the direct-MFMA population-conv path (popconv) was
+        # REMOVED after losing both of its target configs on hardware —
+        # docs/popconv_postmortem.md has the measurements and reasoning.
 
     # ------------------------------------------------------------ internals
     def _loss(self, params_tuple: Tuple[Tensor, ...], x: Tensor, y: Tensor,
@@ -195,10 +181,6 @@ class FusedEngine:
                 Y[i] = target_tfms[i](Y[i])
             return Y
 
-        if self._pop_fn is not None:
-            return self._pop_round(theta, clients, steps_data, local_steps,
-                                   lr, out, clamp_hi, grad_sign, fix_targets)
-
         fedsgd = local_steps == 1
         if fedsgd:
             grad_fn = self._build_grad_fn(shared_params=True)
@@ -240,76 +222,6 @@ class FusedEngine:
                     torch._foreach_add_(views, gl)
                 else:
                     torch._foreach_add_(views, list(grads), alpha=-lr)
-        slab.sub_(theta.unsqueeze(0))
-        if out is not None:
-            out.copy_(slab)
-            return out
-        return slab
-
-    # --------------------------------------------------- population path
-    def _pop_round(self, theta: Tensor, clients, steps_data, local_steps: int,
-                   lr, out: Optional[Tensor], clamp_hi: Tensor,
-                   grad_sign: Tensor, fix_targets) -> Tensor:
-        """Direct-MFMA population training (engine/popmodel.py): explicit
-        autograd over batched [C, *shape] parameters instead of vmap.
-
-        FedSGD (shared θ): parameters enter the graph as stride-0 expands of
-        the flat θ views — popconv reads them broadcast, autograd.grad
-        w.r.t. the EXPANDED tensors returns dense per-client gradients.
-        FedAvg: parameters are detached slab views updated in place.
-        """
-        import torch.nn.functional as F
-
-        C = len(clients)
-        fedsgd = local_steps == 1
-        names = self.spec.names
-
-        def per_client_loss_sum(params, X, Y):
-            logits = self._pop_fn(params, X)  # [C, B, nc]
-            nc = logits.shape[-1]
-            losses = F.cross_entropy(logits.reshape(-1, nc), Y.reshape(-1),
-                                     reduction="none").view(C, -1).mean(1)
-            losses = torch.minimum(torch.clamp(losses, min=0.0), clamp_hi)
-            return (losses * grad_sign).sum()
-
-        if fedsgd:
-            base = dict(self.spec.named_slices(theta))
-            params = {
-                n: base[n].unsqueeze(0).expand(C, *base[n].shape)
-                .detach().requires_grad_() for n in names
-            }
-            X, Y = steps_data[0]
-            loss = per_client_loss_sum(params, X, fix_targets(Y))
-            grads = torch.autograd.grad(loss, [params[n] for n in names])
-            U = out if out is not None else torch.empty(
-                (C, self.spec.d), device=self.device)
-            views = self.spec.batched_views(U)
-            with torch.no_grad():
-                for n, g in zip(names, grads):
-                    views[n].copy_(g)
-            U.mul_(-lr)
-            return U
-
-        if (getattr(self, "_slab", None) is None
-                or self._slab.shape[0] != C):
-            self._slab = torch.empty(C, self.spec.d, device=self.device)
-        slab = self._slab
-        slab.copy_(theta.unsqueeze(0).expand(C, -1))
-        slab_views = self.spec.batched_views(slab)
-        for s in range(local_steps):
-            params = {n: slab_views[n].detach().requires_grad_()
-                      for n in names}
-            X, Y = steps_data[s]
-            loss = per_client_loss_sum(params, X, fix_targets(Y))
-            grads = torch.autograd.grad(loss, [params[n] for n in names])
-            with torch.no_grad():
-                gl = list(grads)
-                if isinstance(lr, torch.Tensor):
-                    torch._foreach_mul_(gl, -lr)
-                    torch._foreach_add_([slab_views[n] for n in names], gl)
-                else:
-                    torch._foreach_add_([slab_views[n] for n in names], gl,
-                                        alpha=-lr)
         slab.sub_(theta.unsqueeze(0))
         if out is not None:
             out.copy_(slab)

@@ -55,3 +55,8 @@ torch.cuda.synchronize()
 print("step8 capture+replay a2a/allreduce ok", flush=True)
 dist.destroy_process_group()
 print("step9 destroy ok — ALL OK", flush=True)
+# interpreter teardown after an NCCL process group can deadlock in some
+# parent environments (observed: clean exit standalone, hang at exit under
+# a pytest parent — gpurun_out/r2_call6.log); the probe's work is done and
+# verified, so skip teardown.
+os._exit(0)

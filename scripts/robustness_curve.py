@@ -34,7 +34,7 @@ def run_one(aggregator, rounds, z, seed=7, clients=100, byz=20,
                             seed=0, device=dev, learnable="templates")
     attack_kws = {"num_clients": clients, "num_byzantine": byz}
     if z is not None:
-        attack_kws = {"z": z}
+        attack_kws = {"num_clients": clients, "num_byzantine": byz, "z": z}
     agg_kws = {}
     if aggregator == "trimmedmean":
         agg_kws = {"nb": byz}

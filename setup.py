@@ -26,14 +26,6 @@ setup(
                 "nvcc": ["-O3", "-std=c++17"],
             },
         ),
-        CUDAExtension(
-            name="blades_amd._hip_popconv",
-            sources=["csrc/popconv.hip"],
-            extra_compile_args={
-                "cxx": ["-O3"],
-                "nvcc": ["-O3", "-std=c++17"],
-            },
-        )
     ],
     cmdclass={"build_ext": BuildExtension},
 )

@@ -19,7 +19,7 @@ def _train(aggregator, rounds=40, z=4.0, seed=7):
                             seed=0, device="cuda:0", learnable="templates")
     agg_kws = {"nb": 20} if aggregator == "trimmedmean" else {}
     sim = Simulator(ds, num_byzantine=20, attack="alie",
-                    attack_kws={"z": z},
+                    attack_kws={"num_clients": 100, "num_byzantine": 20, "z": z},
                     aggregator=aggregator, aggregator_kws=agg_kws,
                     use_cuda=True, log_path=f"/tmp/robust_t_{aggregator}",
                     seed=seed)
