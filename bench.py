@@ -174,6 +174,16 @@ def main():
     phases = {k.replace("blades/", ""): round(v, 3)
               for k, v in sorted(phase_seconds.items())}
 
+    mem = {}
+    if use_cuda:
+        free_b, total_b = torch.cuda.mem_get_info(device)
+        mem = {"max_allocated_gb": round(
+                   torch.cuda.max_memory_allocated(device) / 1e9, 2),
+               "reserved_gb": round(
+                   torch.cuda.memory_reserved(device) / 1e9, 2),
+               "device_total_gb": round(total_b / 1e9, 2),
+               "device_used_gb": round((total_b - free_b) / 1e9, 2)}
+
     if rank == 0:
         print(json.dumps({
             "metric": f"global rounds/sec ({args.model} FedSGD, "
@@ -202,6 +212,7 @@ def main():
                 "parallelism": f"client-sharded dp{n_gpus}",
             },
             "phase_seconds_total": phases,
+            "memory": mem,
         }))
 
 

@@ -82,11 +82,12 @@ class CapturedRound:
                 return "trimmedmean b out of kernel range"
         if not sim.server._plain_sgd():
             return "server optimizer is not plain SGD"
+        if sim._stream_clients:
+            return ("streamed rounds are not captured (a whole-shard "
+                    "capture would defeat the memory point of streaming)")
         if sim.runtime.distributed:
             if dist.get_backend() != "nccl":
                 return "multi-rank capture needs the RCCL backend"
-            if sim._stream_clients:
-                return "streamed coordinate rounds are not captured"
             if sim._use_coordinate() and not getattr(
                     sim.aggregator, "coordinate_shardable", False) \
                     and not isinstance(sim.aggregator, Krum):

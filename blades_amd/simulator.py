@@ -420,6 +420,12 @@ class Simulator:
             fusable, custom = split_fusable(shard)
             if self._engine_choice == "loop":
                 fusable, custom = [], shard
+            # single-rank / full-gather streaming: train the shard in
+            # client chunks written straight into the slab, so transient
+            # vmap gradient outputs are chunk-sized, not shard-sized
+            # (config 5 at ws=1: the [1250, 36.5M] slab alone is 182 GB —
+            # whole-shard grads would double it past 288 GB)
+            stream = self._stream_clients
             # buffer reused across eager rounds (pads were zeroed once and
             # are never written; the engine overwrites [:, :d] fully)
             cache_key = (len(shard), d_pad)
@@ -431,8 +437,15 @@ class Simulator:
             U_local = buf_local[:, :d]
             local_pos = {c.id(): i for i, c in enumerate(shard)}
             if len(fusable) == len(shard):
-                self._fused.run_round(theta, shard, self.dataset,
-                                      local_steps, lr, out=U_local)
+                if stream and stream < len(shard):
+                    for lo in range(0, len(shard), stream):
+                        part = shard[lo:lo + stream]
+                        self._fused.run_round(theta, part, self.dataset,
+                                              local_steps, lr,
+                                              out=U_local[lo:lo + len(part)])
+                else:
+                    self._fused.run_round(theta, shard, self.dataset,
+                                          local_steps, lr, out=U_local)
             else:
                 if fusable:
                     Uf = self._fused.run_round(theta, fusable, self.dataset,

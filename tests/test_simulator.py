@@ -190,3 +190,27 @@ def test_nan_update_is_sanitized(tmp_path):
     sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
     theta = sim.server.flat_parameters()
     assert torch.isfinite(theta).all()
+
+
+def test_single_rank_streamed_matches_unstreamed(tmp_path):
+    """ws=1 client-chunk streaming (config-5 memory mode) is numerically
+    identical to the whole-shard round."""
+    import torch
+
+    from blades_amd import Simulator
+    from blades_amd.datasets import SyntheticFLDataset
+    from blades_amd.models import MLP
+
+    def run(stream):
+        ds = SyntheticFLDataset(num_clients=7, samples_per_client=16,
+                                batch_size=8, shape=(1, 28, 28),
+                                num_classes=10, seed=0)
+        sim = Simulator(ds, num_byzantine=2, attack="signflipping",
+                        aggregator="geomed",
+                        log_path=str(tmp_path / f"s{stream}"), seed=3,
+                        stream_clients=stream)
+        torch.manual_seed(3)
+        sim.run(MLP(), global_rounds=2, validate_interval=0, client_lr=0.1)
+        return sim.server.flat_parameters()
+
+    assert torch.equal(run(None), run(3))
