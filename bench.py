@@ -118,7 +118,12 @@ def main():
     elif args.attack == "labelflipping":
         attack_kws = {"num_classes": args.num_classes}
 
+    # 1-GPU rehearsal of multi-rank launches: force every rank onto one
+    # device (e.g. BLADES_BENCH_DEVICE=cuda:0 with BLADES_AMD_BACKEND=gloo)
+    device_override = os.environ.get("BLADES_BENCH_DEVICE")
+
     sim = Simulator(
+        device=device_override,
         dataset=SyntheticFLDataset(
             num_clients=total_clients,
             samples_per_client=args.samples_per_client,

@@ -23,7 +23,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def run_one(aggregator, rounds, z, seed=7, clients=100, byz=20,
-            eval_every=25):
+            eval_every=25, attack="alie", eps=None):
     from blades_amd import Simulator
     from blades_amd.datasets import SyntheticFLDataset
     from blades_amd.models import resnet18
@@ -35,14 +35,16 @@ def run_one(aggregator, rounds, z, seed=7, clients=100, byz=20,
     attack_kws = {"num_clients": clients, "num_byzantine": byz}
     if z is not None:
         attack_kws = {"num_clients": clients, "num_byzantine": byz, "z": z}
+    if attack == "ipm":
+        attack_kws = {"epsilon": 10.0 if eps is None else eps}
     agg_kws = {}
     if aggregator == "trimmedmean":
         agg_kws = {"nb": byz}
-    sim = Simulator(ds, num_byzantine=byz, attack="alie",
+    sim = Simulator(ds, num_byzantine=byz, attack=attack,
                     attack_kws=attack_kws,
                     aggregator=aggregator, aggregator_kws=agg_kws,
                     use_cuda=dev != "cpu", device=dev,
-                    log_path=f"/tmp/robust_{aggregator}", seed=seed)
+                    log_path=f"/tmp/robust_{attack}_{aggregator}", seed=seed)
     model = resnet18(norm="batch-local")
     torch.manual_seed(seed)
     curve = []
@@ -67,17 +69,23 @@ def main():
                     help="explicit ALIE z (default: paper z_max from n,m)")
     ap.add_argument("--aggregators", type=str,
                     default="mean,trimmedmean,median")
+    ap.add_argument("--attack", type=str, default="alie",
+                    choices=["alie", "ipm", "noise", "signflipping",
+                             "labelflipping"])
+    ap.add_argument("--eps", type=float, default=None, help="IPM epsilon")
     args = ap.parse_args()
     z = None if args.z == "none" else float(args.z)
 
     finals = {}
     curves = {}
     for agg in args.aggregators.split(","):
-        c = run_one(agg, args.rounds, z, eval_every=args.eval_every)
+        c = run_one(agg, args.rounds, z, eval_every=args.eval_every,
+                    attack=args.attack, eps=args.eps)
         curves[agg] = c
         finals[agg] = c[-1]["top1"]
 
-    print("\n| aggregator | final top1 @ round %d |" % args.rounds)
+    print("\n| aggregator | final top1 @ round %d (%s) |"
+          % (args.rounds, args.attack))
     print("|---|---|")
     for agg, t in finals.items():
         print(f"| {agg} | {t:.3f} |")

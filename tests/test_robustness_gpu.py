@@ -1,15 +1,26 @@
-"""End-to-end robustness at CIFAR scale: with 20%
-ALIE attackers on learnable template data, TrimmedMean keeps the accuracy
-that plain Mean loses.  The full slow curve (paper z, 300 rounds) lives in
-scripts/robustness_curve.py + docs; this test uses an aggressive explicit
-z so the separation appears within ~40 rounds."""
+"""End-to-end robustness at CIFAR scale.
+
+Demonstrates the simulator's reason to exist on learnable template data
+(ResNet-18, 100 clients, 20% attackers):
+
+* under IPM (inner-product manipulation, large ε) plain Mean is driven by
+  −ε·mean(honest) into gradient ascent and never learns, while
+  TrimmedMean trims the manipulated rows and trains to high accuracy;
+* the converse is ALSO real and documented: ALIE at its paper z sits
+  inside the honest distribution and *defeats* coordinate-wise robust
+  aggregators while barely perturbing Mean (measured 300-round curves in
+  docs/robustness.md / gpurun_out/r2_call9.log — consistent with
+  Baruch et al. 2019 and the Blades paper's own findings).
+
+top1 here is in PERCENT (reference metric scale).
+"""
 import pytest
 import torch
 
 pytestmark = pytest.mark.gpu
 
 
-def _train(aggregator, rounds=40, z=4.0, seed=7):
+def _train(aggregator, attack, attack_kws, rounds=40, seed=7):
     from blades_amd import Simulator
     from blades_amd.datasets import SyntheticFLDataset
     from blades_amd.models import resnet18
@@ -18,10 +29,11 @@ def _train(aggregator, rounds=40, z=4.0, seed=7):
                             batch_size=32, shape=(3, 32, 32), num_classes=10,
                             seed=0, device="cuda:0", learnable="templates")
     agg_kws = {"nb": 20} if aggregator == "trimmedmean" else {}
-    sim = Simulator(ds, num_byzantine=20, attack="alie",
-                    attack_kws={"num_clients": 100, "num_byzantine": 20, "z": z},
+    sim = Simulator(ds, num_byzantine=20, attack=attack,
+                    attack_kws=attack_kws,
                     aggregator=aggregator, aggregator_kws=agg_kws,
-                    use_cuda=True, log_path=f"/tmp/robust_t_{aggregator}",
+                    use_cuda=True,
+                    log_path=f"/tmp/robust_t_{attack}_{aggregator}",
                     seed=seed)
     torch.manual_seed(seed)
     sim.run(resnet18(norm="batch-local"), global_rounds=rounds, local_steps=1,
@@ -31,10 +43,9 @@ def _train(aggregator, rounds=40, z=4.0, seed=7):
 
 
 @pytest.mark.timeout(900)
-def test_trimmedmean_recovers_what_mean_loses_under_alie():
-    top1_tm = _train("trimmedmean")
-    top1_mean = _train("mean")
-    # template data is easy: a robust aggregator should be well into
-    # learning by round 40 while the attacked mean lags behind
-    assert top1_tm > 0.6, f"trimmedmean failed to learn: {top1_tm}"
-    assert top1_tm > top1_mean + 0.15, (top1_tm, top1_mean)
+def test_trimmedmean_recovers_what_mean_loses_under_ipm():
+    kws = {"epsilon": 10.0}
+    top1_tm = _train("trimmedmean", "ipm", kws)
+    top1_mean = _train("mean", "ipm", kws)
+    assert top1_tm > 60.0, f"trimmedmean failed to learn: {top1_tm}"
+    assert top1_tm > top1_mean + 20.0, (top1_tm, top1_mean)
