@@ -70,7 +70,8 @@ def test_masked_col_mean_std(ext, K, d):
 
 @pytest.mark.parametrize("K,b", [(9, 2), (100, 20), (100, 49), (64, 31),
                                  (7, 0), (20, 8), (70, 32), (33, 16),
-                                 (1000, 20), (12, 5)])
+                                 (1000, 20), (12, 5), (1000, 499),
+                                 (1280, 500), (1300, 500)])
 def test_trimmed_mean(ext, K, b):
     from blades_amd.ops import torch_ref
     U = randU(K, 50000, seed=b)
