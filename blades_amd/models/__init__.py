@@ -12,6 +12,9 @@ from torch import nn
 from .mlp import MLP
 from .resnet import ResNet, WideResNet, resnet18, wide_resnet28_10
 from .cct import CCT, CCTNet, cct_2_3x2_32, cvt_2_4_32, vit_lite_2_4_32
+from .text_cct import (TextCCT, text_cct_2, text_cct_4, text_cct_6,
+                       text_transformer_2, text_transformer_4,
+                       text_transformer_6)
 
 _REGISTRY = {
     "mlp": lambda **kw: MLP(**kw),
@@ -20,6 +23,10 @@ _REGISTRY = {
     "wideresnet28_10": lambda **kw: wide_resnet28_10(**kw),
     "cct": lambda **kw: cct_2_3x2_32(**kw),
     "cct_2_3x2_32": lambda **kw: cct_2_3x2_32(**kw),
+    "text_cct_2": lambda **kw: text_cct_2(**kw),
+    "text_cct_4": lambda **kw: text_cct_4(**kw),
+    "text_cct_6": lambda **kw: text_cct_6(**kw),
+    "text_transformer_2": lambda **kw: text_transformer_2(**kw),
     "cvt": lambda **kw: cvt_2_4_32(**kw),
     "vit_lite": lambda **kw: vit_lite_2_4_32(**kw),
 }

@@ -65,3 +65,28 @@ def test_backward_runs():
         loss.backward()
         assert all(p.grad is not None for p in m.parameters()
                    if p.requires_grad)
+
+
+def test_text_cct_variants():
+    """Masked text models (reference cctnets/text/): forward+backward with
+    padding masks; masked positions must not affect the logits."""
+    import torch
+
+    from blades_amd.models.text_cct import (text_cct_2, text_transformer_2)
+
+    for fn in (text_cct_2, text_transformer_2):
+        m = fn(seq_len=32, vocab_size=100, num_classes=4,
+               word_embedding_dim=64)
+        m.eval()
+        ids = torch.randint(1, 100, (3, 32))
+        ids[:, 20:] = 0  # padding
+        out = m(ids)
+        assert out.shape == (3, 4)
+        # changing PADDED ids must not change the output
+        ids2 = ids.clone()
+        ids2[:, 25:] = 0  # same mask, same content
+        assert torch.allclose(m(ids2), out, atol=1e-6)
+        m.train()
+        loss = torch.nn.functional.cross_entropy(
+            m(ids), torch.randint(0, 4, (3,)))
+        loss.backward()
