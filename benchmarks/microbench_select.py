@@ -17,8 +17,12 @@ for K, b, d in [(100, 20, 11173962), (100, 49, 11173962), (1000, 499, 11173962),
     U = buf[:, :d]
     r1 = t(lambda: ext.trimmed_mean(U, b), n=10)
     r2 = t(lambda: ext.trimmed_mean_radix(U, b), n=10)
+    r3 = None
+    if K <= 1280 and b > 0:
+        r3 = t(lambda: ext.trimmed_mean_radix_lds(U, b), n=10)
     gb = K*d*4/1e9
-    print(f"K={K} b={b} d={d}: auto {r1:.2f} ms, radix {r2:.2f} ms  (slab {gb:.1f} GB)", flush=True)
+    lds = f", radix-lds {r3:.2f} ms" if r3 is not None else ""
+    print(f"K={K} b={b} d={d}: auto {r1:.2f} ms, radix {r2:.2f} ms{lds}  (slab {gb:.1f} GB)", flush=True)
     ok = torch.allclose(ext.trimmed_mean(U, b), ext.trimmed_mean_radix(U, b), atol=1e-5)
     print("   agree:", ok, flush=True)
     del U, buf

@@ -1064,8 +1064,12 @@ static torch::Tensor trimmed_core(torch::Tensor U, long b_lo, long b_hi,
           (int)b_hi, inv);
     return out;
   }
-  // mid K: LDS-staged dual radix — one HBM pass instead of five
-  if (v.K <= RT_L_KMAX) {
+  // NOTE: an LDS-staged one-HBM-pass variant was measured SLOWER than
+  // the 5-pass kernel at its target shapes (549 vs 207 ms at K=1000,
+  // b=499, d=11.2M — 1 workgroup/CU occupancy + same-bin LDS-atomic
+  // serialization outweigh the 5x HBM saving; gpurun_out/r2_call11.log).
+  // It remains reachable via trimmed_mean_radix_lds for benchmarking.
+  if (false && v.K <= RT_L_KMAX) {
     const size_t smem = (size_t)RT_L * 8 + 6 * RT_L * 4
         + 2 * (size_t)RT_L * 256 * 4 + (size_t)v.K * RT_L * 4;
     static bool attr_set = false;
@@ -1114,6 +1118,25 @@ torch::Tensor col_trimmed_sum(torch::Tensor U, long b_lo, long b_hi) {
 
 torch::Tensor trimmed_mean(torch::Tensor U, long b) {
   return trimmed_mean_select(U, b);
+}
+
+torch::Tensor trimmed_mean_radix_lds(torch::Tensor U, long b) {
+  // benchmark-only entry for the LDS-staged radix (see note in
+  // trimmed_core: measured slower than the 5-pass kernel)
+  auto v = view_of(U);
+  TORCH_CHECK(v.K - 2 * b >= 1 && b > 0 && v.K <= RT_L_KMAX, "bad shape");
+  auto out = torch::empty({v.d}, U.options());
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  const size_t smem = (size_t)RT_L * 8 + 6 * RT_L * 4
+      + 2 * (size_t)RT_L * 256 * 4 + (size_t)v.K * RT_L * 4;
+  hipFuncSetAttribute(
+      reinterpret_cast<const void*>(&radix_trimmed_lds_kernel),
+      hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+  const int grid = (int)std::min<long long>(cdiv(v.d, RT_L), kMaxBlocks);
+  radix_trimmed_lds_kernel<<<grid, 256, smem, stream>>>(
+      v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, b, b,
+      1.0 / (double)(v.K - 2 * b));
+  return out;
 }
 
 torch::Tensor trimmed_mean_radix(torch::Tensor U, long b) {
@@ -1212,6 +1235,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("trimmed_mean", &trimmed_mean, "coordinate-wise trimmed mean (K3)");
   m.def("col_trimmed_sum", &col_trimmed_sum,
         "asymmetric trimmed column sum (K5 krum scores)");
+  m.def("trimmed_mean_radix_lds", &trimmed_mean_radix_lds,
+        "LDS-staged radix (benchmark-only; slower, kept for evidence)");
   m.def("trimmed_mean_radix", &trimmed_mean_radix,
         "trimmed mean via dual radix select (benchmarking entry)");
   m.def("col_median", &col_median, "coordinate-wise median (K2)");
