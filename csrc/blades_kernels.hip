@@ -1046,8 +1046,9 @@ static torch::Tensor trimmed_core(torch::Tensor U, long b_lo, long b_hi,
   const long bmax = std::max(b_lo, b_hi);
   // small trims: divergence-free register-selection kernel (headline
   // path).  NB = smallest capacity >= bmax so a b=2 median does not pay
-  // the NB=32 sort network.
-  if (bmax <= 32) {
+  // a bigger sort network.  NB=64 (2 waves/SIMD at ~210 VGPRs) covers
+  // the K<=129 median that the LDS-heap kernel served divergence-bound.
+  if (bmax <= 64) {
     const int BS = 256;
     const int grid = col_grid(v.d, BS);
     if (bmax <= 8)
@@ -1058,8 +1059,12 @@ static torch::Tensor trimmed_core(torch::Tensor U, long b_lo, long b_hi,
       trimmed_regsel_kernel<16><<<grid, BS, 0, stream>>>(
           v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo,
           (int)b_hi, inv);
-    else
+    else if (bmax <= 32)
       trimmed_regsel_kernel<32><<<grid, BS, 0, stream>>>(
+          v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo,
+          (int)b_hi, inv);
+    else
+      trimmed_regsel_kernel<64><<<grid, BS, 0, stream>>>(
           v.ptr, out.data_ptr<float>(), v.K, v.d, v.ld, (int)b_lo,
           (int)b_hi, inv);
     return out;

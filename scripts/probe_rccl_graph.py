@@ -53,10 +53,9 @@ with torch.cuda.graph(g2):
 g2.replay()
 torch.cuda.synchronize()
 print("step8 capture+replay a2a/allreduce ok", flush=True)
-dist.destroy_process_group()
-print("step9 destroy ok — ALL OK", flush=True)
-# interpreter teardown after an NCCL process group can deadlock in some
-# parent environments (observed: clean exit standalone, hang at exit under
-# a pytest parent — gpurun_out/r2_call6.log); the probe's work is done and
-# verified, so skip teardown.
+print("step9 ALL OK", flush=True)
+# dist.destroy_process_group() deadlocks under a pytest parent (observed:
+# step8 prints, destroy never returns — gpurun_out/r2_call13.log; clean
+# standalone).  The probe's work is done and verified; exit hard and let
+# the OS reclaim the communicator.
 os._exit(0)
