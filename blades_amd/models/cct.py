@@ -248,3 +248,47 @@ class CCTNet(nn.Module):
 
 def create_model():
     return CCTNet(), nn.CrossEntropyLoss()
+
+
+# ------------------------------------------------------------- checkpoints
+def resize_positional_embedding(pe: torch.Tensor,
+                                target_len: int) -> torch.Tensor:
+    """Bilinearly rescale a [1, N, D] learnable positional embedding to a
+    new sequence length (square token grids), so checkpoints trained at
+    one image size load at another (the capability of the reference's
+    pretrained-URL path, cctnets/utils/helpers.py resize_pos_embed)."""
+    import math
+
+    n, d = pe.shape[1], pe.shape[2]
+    if n == target_len:
+        return pe
+    gs_old = int(math.sqrt(n))
+    gs_new = int(math.sqrt(target_len))
+    if gs_old * gs_old != n or gs_new * gs_new != target_len:
+        raise ValueError(f"non-square token grids: {n} -> {target_len}")
+    grid = pe.reshape(1, gs_old, gs_old, d).permute(0, 3, 1, 2)
+    grid = F.interpolate(grid, size=(gs_new, gs_new), mode="bilinear",
+                         align_corners=False)
+    return grid.permute(0, 2, 3, 1).reshape(1, target_len, d)
+
+
+def load_pretrained(model: "CCT", path: str, strict: bool = True) -> "CCT":
+    """Load a LOCAL state-dict file into a CCT-family model, resizing the
+    positional embedding and dropping a mismatched classifier head.
+
+    This environment has no egress, so the reference's
+    load_state_dict_from_url path (cctnets/cct.py:106-117) maps to
+    loading an already-downloaded file; the adaptation semantics
+    (pos-embed resize, head reset on class-count change) are preserved.
+    """
+    sd = torch.load(path, map_location="cpu", weights_only=True)
+    own = model.state_dict()
+    pe_key = "positional_emb"
+    if pe_key in sd and pe_key in own and sd[pe_key].shape != own[pe_key].shape:
+        sd[pe_key] = resize_positional_embedding(sd[pe_key],
+                                                 own[pe_key].shape[1])
+    for key in ("fc.weight", "fc.bias"):
+        if key in sd and key in own and sd[key].shape != own[key].shape:
+            sd[key] = own[key]  # class count changed: keep the fresh head
+    model.load_state_dict(sd, strict=strict)
+    return model

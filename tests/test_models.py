@@ -90,3 +90,26 @@ def test_text_cct_variants():
         loss = torch.nn.functional.cross_entropy(
             m(ids), torch.randint(0, 4, (3,)))
         loss.backward()
+
+
+def test_cct_load_pretrained_resize(tmp_path):
+    """Local-checkpoint loading with pos-embed resize + head reset
+    (reference: cctnets pretrained-URL path, no egress here)."""
+    import torch
+
+    from blades_amd.models.cct import CCT, load_pretrained
+
+    src = CCT(img_size=32, num_classes=10)
+    p = str(tmp_path / "cct.pt")
+    torch.save(src.state_dict(), p)
+
+    # different image size (pos-embed resize) AND class count (head reset)
+    dst = CCT(img_size=64, num_classes=5)
+    load_pretrained(dst, p)
+    out = dst(torch.randn(2, 3, 64, 64))
+    assert out.shape == (2, 5)
+    # same-shape load stays bitwise
+    dst2 = CCT(img_size=32, num_classes=10)
+    load_pretrained(dst2, p)
+    assert torch.equal(dst2.state_dict()["fc.weight"],
+                       src.state_dict()["fc.weight"])
