@@ -32,11 +32,18 @@ def run_one(aggregator, rounds, z, seed=7, clients=100, byz=20,
     ds = SyntheticFLDataset(num_clients=clients, samples_per_client=64,
                             batch_size=32, shape=(3, 32, 32), num_classes=10,
                             seed=0, device=dev, learnable="templates")
-    attack_kws = {"num_clients": clients, "num_byzantine": byz}
-    if z is not None:
-        attack_kws = {"num_clients": clients, "num_byzantine": byz, "z": z}
-    if attack == "ipm":
+    if attack == "alie":
+        attack_kws = {"num_clients": clients, "num_byzantine": byz}
+        if z is not None:
+            attack_kws["z"] = z
+    elif attack == "ipm":
         attack_kws = {"epsilon": 10.0 if eps is None else eps}
+    elif attack == "noise":
+        attack_kws = {"mean": 0.1, "std": 10.0}
+    elif attack == "labelflipping":
+        attack_kws = {"num_classes": 10}
+    else:
+        attack_kws = {}
     agg_kws = {}
     if aggregator == "trimmedmean":
         agg_kws = {"nb": byz}
