@@ -16,6 +16,7 @@ EXAMPLES = os.path.join(REPO, "examples")
     "customize_attack.py",
     "plot_comparing_aggregation_schemes.py",
     "simulation_sweep.py",
+    "robustness_comparison.py",
 ])
 @pytest.mark.timeout(300)
 def test_example_runs(script, tmp_path):
