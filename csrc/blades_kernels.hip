@@ -924,6 +924,72 @@ __global__ void gram_mfma_kernel(const float* __restrict__ U,
   }
 }
 
+// 64x64 tile-pair variant: halves HBM traffic (each block covers 4x the
+// output area per staged chunk) and each wave OWNS one 32x32 output
+// sub-tile — 4x the MFMAs per stage, no cross-wave LDS combine.
+__global__ __launch_bounds__(256)
+void gram_mfma64_kernel(const float* __restrict__ U,
+                        float* __restrict__ G,
+                        long long K, long long d, long long ld,
+                        int ntiles, long long slice_len) {
+  __shared__ float As[64][GRAM_BD + 1];
+  __shared__ float Bs[64][GRAM_BD + 1];
+
+  const int pair = blockIdx.x;
+  const int it = pair / ntiles;
+  const int jt = pair % ntiles;
+  if (jt < it) return;  // symmetric: upper triangle only, mirrored below
+
+  const long long c_begin = (long long)blockIdx.y * slice_len;
+  const long long c_end = (c_begin + slice_len < d) ? c_begin + slice_len : d;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int sub_i = wid >> 1;   // wave's A-row block (0/1)
+  const int sub_j = wid & 1;    // wave's B-row block (0/1)
+
+  f32x16 acc = {};
+
+  for (long long c0 = c_begin; c0 < c_end; c0 += GRAM_BD) {
+    const int cols = (int)min((long long)GRAM_BD, c_end - c0);
+    __syncthreads();
+    for (int e = threadIdx.x; e < 64 * GRAM_BD; e += blockDim.x) {
+      const int r = e / GRAM_BD, cc = e % GRAM_BD;
+      const long long gi = (long long)it * 64 + r;
+      const long long gj = (long long)jt * 64 + r;
+      As[r][cc] = (gi < K && cc < cols) ? U[gi * ld + c0 + cc] : 0.f;
+      Bs[r][cc] = (gj < K && cc < cols) ? U[gj * ld + c0 + cc] : 0.f;
+    }
+    __syncthreads();
+    const int row = lane & 31;
+    const int ksel = lane >> 5;  // which of the 2 contraction columns
+    #pragma unroll 4
+    for (int s = 0; s < GRAM_BD / 2; ++s) {
+      const int c = 2 * s + ksel;
+      const float a = As[32 * sub_i + row][c];
+      const float b = Bs[32 * sub_j + row][c];
+      acc = __builtin_amdgcn_mfma_f32_32x32x2f32(a, b, acc, 0, 0, 0);
+    }
+  }
+
+  // each wave writes its own 32x32 sub-tile straight from the
+  // accumulator (atomic across blockIdx.y slices)
+  {
+    const int col = lane & 31;
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int rw = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+      const long long gi = (long long)it * 64 + 32 * sub_i + rw;
+      const long long gj = (long long)jt * 64 + 32 * sub_j + col;
+      if (gi < K && gj < K) {
+        atomicAdd(&G[gi * K + gj], acc[r]);
+        if (it != jt)
+          atomicAdd(&G[gj * K + gi], acc[r]);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // host wrappers
 // ---------------------------------------------------------------------------
@@ -1209,7 +1275,7 @@ torch::Tensor row_dots(torch::Tensor U, torch::Tensor vv) {
   return row_reduce(U, vc.data_ptr<float>(), 2);
 }
 
-torch::Tensor gram(torch::Tensor U) {
+static torch::Tensor gram_t32(torch::Tensor U) {
   auto v = view_of(U);
   auto G = torch::zeros({v.K, v.K}, U.options());
   const int ntiles = (int)cdiv(v.K, 32);
@@ -1224,6 +1290,29 @@ torch::Tensor gram(torch::Tensor U) {
   gram_mfma_kernel<<<grid, 256, 0, stream>>>(
       v.ptr, G.data_ptr<float>(), v.K, v.d, v.ld, ntiles, slice);
   return G;
+}
+
+static torch::Tensor gram_t64(torch::Tensor U) {
+  auto v = view_of(U);
+  auto G = torch::zeros({v.K, v.K}, U.options());
+  const int ntiles = (int)cdiv(v.K, 64);
+  const int npairs = ntiles * ntiles;
+  long long nsl = std::max<long long>(1, 1024 / std::max(1, npairs));
+  long long slice = std::max<long long>(GRAM_BD, cdiv(v.d, nsl));
+  slice = cdiv(slice, GRAM_BD) * GRAM_BD;
+  nsl = cdiv(v.d, slice);
+  dim3 grid((unsigned)npairs, (unsigned)nsl);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  gram_mfma64_kernel<<<grid, 256, 0, stream>>>(
+      v.ptr, G.data_ptr<float>(), v.K, v.d, v.ld, ntiles, slice);
+  return G;
+}
+
+torch::Tensor gram(torch::Tensor U) {
+  // 64x64 tile pairs halve the HBM traffic (K²d/64 vs K²d/32 reads) and
+  // give each wave 4x the MFMAs per LDS stage; below ~3 tiles the small
+  // kernel fills the chip better.
+  return (U.size(0) >= 96) ? gram_t64(U) : gram_t32(U);
 }
 
 }  // namespace
