@@ -68,3 +68,46 @@ def test_resume_reproduces_stateful_aggregator(tmp_path):
     theta_c = sim_c.server.flat_parameters()
 
     assert torch.allclose(theta_a, theta_c, atol=1e-7)
+
+
+def test_centeredclipping_momentum_relayout(tmp_path):
+    """Momentum carried across gather layouts / world sizes is re-laid
+    out (trim/pad), not silently zeroed (round-1 advisor finding)."""
+    import torch
+
+    from blades_amd.aggregators import Centeredclipping
+
+    class _FakeRuntime:
+        world_size = 2
+        rank = 0
+        distributed = True
+
+        def all_reduce_(self, t, op="sum"):
+            return t  # single-process stand-in: partials are full sums
+
+        def all_gather_flat(self, shard):
+            # rank 0's shard + zeros for the absent rank (layout test only)
+            return torch.cat([shard, torch.zeros_like(shard)])
+
+    agg = Centeredclipping(tau=10.0, n_iter=2)
+    U = torch.randn(6, 8)
+    agg(U)  # momentum is now length 8 (full form)
+    m_full = agg.momentum.clone()
+
+    # switch to a padded coordinate-shard form: dshard*ws = 10 > 8
+    U_shard = torch.randn(6, 5)
+    agg.aggregate_shard(U_shard, _FakeRuntime())
+    # the first 8 momentum coords must have been CARRIED (padded form
+    # starts from m_full, not zeros): verify by reconstructing the
+    # expected first-iteration shard update from the carried slice
+    assert agg.momentum.numel() == 10
+
+    agg2 = Centeredclipping(tau=10.0, n_iter=2)
+    agg2.momentum = m_full.clone()
+    carried = torch.cat([m_full, torch.zeros(2)])
+    agg3 = Centeredclipping(tau=10.0, n_iter=2)
+    agg3.momentum = carried.clone()
+    # same shard aggregation from the explicitly padded state must match
+    out2 = agg2.aggregate_shard(U_shard, _FakeRuntime())
+    out3 = agg3.aggregate_shard(U_shard, _FakeRuntime())
+    assert torch.allclose(out2, out3, atol=1e-6)
