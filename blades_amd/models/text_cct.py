@@ -61,10 +61,17 @@ class TextTokenizer(nn.Module):
         # x: [B, N, E] -> conv over N
         y = self.pool(F.relu(self.conv(x.transpose(1, 2)))).transpose(1, 2)
         if mask is not None:
-            m = self.pool(self.conv(mask.float().unsqueeze(1).expand(
-                -1, self.conv.in_channels, -1)).abs()).transpose(1, 2)
-            mask = m.amax(dim=2) > 0
-            mask = mask[:, :y.shape[1]]
+            # propagate validity through the SAME geometry with a ones
+            # kernel (never the learned weights — near-zero weights would
+            # silently invalidate real positions): a downsampled slot is
+            # valid iff its receptive field saw any valid token
+            ones = torch.ones(1, 1, self.conv.kernel_size[0],
+                              device=mask.device)
+            m = F.conv1d(mask.float().unsqueeze(1), ones,
+                         stride=self.conv.stride[0],
+                         padding=self.conv.padding[0])
+            m = self.pool(m).squeeze(1)
+            mask = (m > 0)[:, :y.shape[1]]
         return y, mask
 
 
